@@ -1,0 +1,273 @@
+#include "ps.h"
+
+#include <poll.h>
+#include <unistd.h>
+
+#include <cstring>
+
+namespace tma {
+
+namespace {
+
+enum FrameType : uint8_t { kUpdate = 1, kFetch = 2, kAck = 3, kData = 4 };
+
+struct FrameHdr {
+  uint8_t type;
+  uint8_t rule;   // 0 = copy, 1 = add
+  uint8_t dtype;
+  uint8_t pad;
+  int32_t tid;
+  uint64_t seq;
+  uint64_t bytes;  // payload bytes following the header
+};
+static_assert(sizeof(FrameHdr) == 24, "frame header size");
+
+}  // namespace
+
+PsContext::PsContext(int rank, const std::string& bind_addr)
+    : rank_(rank), mesh_(bind_addr) {}
+
+PsContext::~PsContext() { stop(); }
+
+void PsContext::start(int rank, const std::vector<std::string>& addrs) {
+  rank_ = rank;
+  world_ = static_cast<int>(addrs.size());
+  if (world_ > 1) mesh_.connect_mesh(rank, addrs);
+  sock_m_.clear();
+  for (int i = 0; i < world_; ++i) sock_m_.push_back(std::make_unique<std::mutex>());
+  TMA_CHECK(pipe(wake_pipe_) == 0, "pipe failed");
+  running_ = true;
+  if (world_ > 1) loop_thread_ = std::thread([this] { loop(); });
+}
+
+void PsContext::stop() {
+  if (!running_.exchange(false)) return;
+  if (wake_pipe_[1] >= 0) {
+    char b = 1;
+    (void)!write(wake_pipe_[1], &b, 1);
+  }
+  if (loop_thread_.joinable()) loop_thread_.join();
+  for (int i = 0; i < 2; ++i)
+    if (wake_pipe_[i] >= 0) {
+      close(wake_pipe_[i]);
+      wake_pipe_[i] = -1;
+    }
+}
+
+PsContext::Reg& PsContext::reg(int tid) {
+  std::lock_guard<std::mutex> lk(regs_m_);
+  auto it = regs_.find(tid);
+  TMA_CHECK(it != regs_.end(), "parameterserver: unregistered tensor id");
+  return *it->second;
+}
+
+void PsContext::register_tensor(int tid, size_t count, DType d, const void* init) {
+  auto r = std::make_unique<Reg>();
+  r->count = count;
+  r->dtype = d;
+  split_segments(count, world_, r->off, r->cnt);
+  size_t es = dtype_size(d);
+  r->shard.resize(r->cnt[rank_] * es);
+  std::memcpy(r->shard.data(),
+              static_cast<const char*>(init) + r->off[rank_] * es,
+              r->cnt[rank_] * es);
+  std::lock_guard<std::mutex> lk(regs_m_);
+  regs_[tid] = std::move(r);
+}
+
+void PsContext::free_tensor(int tid) {
+  std::lock_guard<std::mutex> lk(regs_m_);
+  regs_.erase(tid);
+}
+
+void PsContext::shard_of(int tid, int r, size_t& off, size_t& cnt) {
+  Reg& g = reg(tid);
+  off = g.off[r];
+  cnt = g.cnt[r];
+}
+
+void PsContext::write_frame(int peer, const void* hdr, size_t hdr_n,
+                            const void* payload, size_t payload_n) {
+  std::lock_guard<std::mutex> lk(*sock_m_[peer]);
+  mesh_.send_all(peer, hdr, hdr_n);
+  if (payload_n > 0) mesh_.send_all(peer, payload, payload_n);
+}
+
+std::shared_ptr<Work> PsContext::send(int tid, const void* data, int rule) {
+  Reg& g = reg(tid);
+  size_t es = dtype_size(g.dtype);
+  auto pend = std::make_shared<Pending>();
+  pend->work = std::make_shared<Work>();
+  uint64_t seq = next_seq_.fetch_add(1);
+  // Count remote owners first so an early ACK cannot complete prematurely.
+  int remotes = 0;
+  for (int r = 0; r < world_; ++r)
+    if (r != rank_ && g.cnt[r] > 0) ++remotes;
+  pend->remaining = remotes;
+  if (remotes > 0) {
+    std::lock_guard<std::mutex> lk(pend_m_);
+    pending_[seq] = pend;
+  }
+  // Local shard: apply directly.
+  if (g.cnt[rank_] > 0) {
+    std::lock_guard<std::mutex> lk(g.m);
+    const char* src = static_cast<const char*>(data) + g.off[rank_] * es;
+    if (rule == 0)
+      std::memcpy(g.shard.data(), src, g.cnt[rank_] * es);
+    else
+      cpu_reduce(g.shard.data(), src, g.cnt[rank_], g.dtype, RedOp::SUM);
+  }
+  // Remote shards.
+  for (int r = 0; r < world_; ++r) {
+    if (r == rank_ || g.cnt[r] == 0) continue;
+    FrameHdr h{kUpdate, static_cast<uint8_t>(rule),
+               static_cast<uint8_t>(static_cast<int>(g.dtype)), 0, tid, seq,
+               g.cnt[r] * es};
+    write_frame(r, &h, sizeof(h), static_cast<const char*>(data) + g.off[r] * es,
+                g.cnt[r] * es);
+  }
+  if (remotes == 0) pend->work->finish(nullptr);
+  return pend->work;
+}
+
+std::shared_ptr<Work> PsContext::receive(int tid, void* data) {
+  Reg& g = reg(tid);
+  size_t es = dtype_size(g.dtype);
+  auto pend = std::make_shared<Pending>();
+  pend->work = std::make_shared<Work>();
+  pend->dest = data;
+  pend->tid = tid;
+  uint64_t seq = next_seq_.fetch_add(1);
+  int remotes = 0;
+  for (int r = 0; r < world_; ++r)
+    if (r != rank_ && g.cnt[r] > 0) ++remotes;
+  pend->remaining = remotes;
+  if (remotes > 0) {
+    std::lock_guard<std::mutex> lk(pend_m_);
+    pending_[seq] = pend;
+  }
+  if (g.cnt[rank_] > 0) {
+    std::lock_guard<std::mutex> lk(g.m);
+    std::memcpy(static_cast<char*>(data) + g.off[rank_] * es, g.shard.data(),
+                g.cnt[rank_] * es);
+  }
+  for (int r = 0; r < world_; ++r) {
+    if (r == rank_ || g.cnt[r] == 0) continue;
+    FrameHdr h{kFetch, 0, 0, 0, tid, seq, 0};
+    write_frame(r, &h, sizeof(h), nullptr, 0);
+  }
+  if (remotes == 0) pend->work->finish(nullptr);
+  return pend->work;
+}
+
+void PsContext::handle_frame(int peer) {
+  FrameHdr h;
+  mesh_.recv_all(peer, &h, sizeof(h));
+  switch (h.type) {
+    case kUpdate: {
+      Reg& g = reg(h.tid);
+      std::vector<char> buf(h.bytes);
+      mesh_.recv_all(peer, buf.data(), h.bytes);
+      {
+        std::lock_guard<std::mutex> lk(g.m);
+        TMA_CHECK(h.bytes == g.shard.size(), "update size mismatch");
+        if (h.rule == 0)
+          std::memcpy(g.shard.data(), buf.data(), h.bytes);
+        else
+          cpu_reduce(g.shard.data(), buf.data(), g.cnt[rank_], g.dtype, RedOp::SUM);
+      }
+      FrameHdr ack{kAck, 0, 0, 0, h.tid, h.seq, 0};
+      write_frame(peer, &ack, sizeof(ack), nullptr, 0);
+      break;
+    }
+    case kFetch: {
+      Reg& g = reg(h.tid);
+      std::vector<char> snapshot;
+      {
+        std::lock_guard<std::mutex> lk(g.m);
+        snapshot = g.shard;
+      }
+      FrameHdr d{kData, 0, static_cast<uint8_t>(static_cast<int>(g.dtype)), 0, h.tid,
+                 h.seq, snapshot.size()};
+      write_frame(peer, &d, sizeof(d), snapshot.data(), snapshot.size());
+      break;
+    }
+    case kAck: {
+      std::shared_ptr<Pending> p;
+      {
+        std::lock_guard<std::mutex> lk(pend_m_);
+        auto it = pending_.find(h.seq);
+        if (it != pending_.end()) p = it->second;
+      }
+      if (p && p->remaining.fetch_sub(1) == 1) {
+        {
+          std::lock_guard<std::mutex> lk(pend_m_);
+          pending_.erase(h.seq);
+        }
+        p->work->finish(nullptr);
+      }
+      break;
+    }
+    case kData: {
+      std::shared_ptr<Pending> p;
+      {
+        std::lock_guard<std::mutex> lk(pend_m_);
+        auto it = pending_.find(h.seq);
+        if (it != pending_.end()) p = it->second;
+      }
+      TMA_CHECK(p != nullptr, "DATA reply with no pending receive");
+      Reg& g = reg(h.tid);
+      size_t es = dtype_size(g.dtype);
+      mesh_.recv_all(peer, static_cast<char*>(p->dest) + g.off[peer] * es, h.bytes);
+      if (p->remaining.fetch_sub(1) == 1) {
+        {
+          std::lock_guard<std::mutex> lk(pend_m_);
+          pending_.erase(h.seq);
+        }
+        p->work->finish(nullptr);
+      }
+      break;
+    }
+    default:
+      TMA_CHECK(false, "bad PS frame type");
+  }
+}
+
+void PsContext::loop() {
+  std::vector<pollfd> fds;
+  std::vector<int> peer_of;
+  while (running_) {
+    fds.clear();
+    peer_of.clear();
+    fds.push_back({wake_pipe_[0], POLLIN, 0});
+    peer_of.push_back(-1);
+    for (int r = 0; r < world_; ++r) {
+      if (r == rank_) continue;
+      // TcpMesh keeps one fd per peer; poll it for incoming frames.
+      fds.push_back({-1, POLLIN, 0});
+      peer_of.push_back(r);
+    }
+    // Fill fds from the mesh (private access via recv path: we poll the
+    // same sockets; safe because this thread is the only reader).
+    for (size_t i = 1; i < fds.size(); ++i) fds[i].fd = mesh_fd(peer_of[i]);
+    int rc = poll(fds.data(), fds.size(), 500);
+    if (rc <= 0) continue;
+    if (fds[0].revents & POLLIN) {
+      char b;
+      (void)!read(wake_pipe_[0], &b, 1);
+      continue;
+    }
+    for (size_t i = 1; i < fds.size(); ++i) {
+      if (fds[i].revents & POLLIN) {
+        try {
+          handle_frame(peer_of[i]);
+        } catch (const std::exception& e) {
+          if (!running_) return;
+          throw;
+        }
+      }
+    }
+  }
+}
+
+}  // namespace tma

@@ -1,0 +1,95 @@
+// Sharded parameter server over the TCP mesh.
+//
+// Re-design of TorchMPI's parameter server (lib/parameterserver.cpp,
+// SURVEY.md sections 2.1 and 3.4): a registered tensor is sharded into
+// world_size contiguous pieces, shard i owned by rank i. Clients push
+// updates with a rule ('copy' = overwrite, 'add' = accumulate) and pull
+// current values asynchronously; integer handles synchronize.
+//
+// Wire design (no MPI): the PS owns its own TCP mesh (separate from the
+// collective mesh so PS traffic never interleaves with collectives). One
+// event-loop thread per rank reads all peer sockets and dispatches frames:
+// requests are served against the local shard store; replies complete
+// pending client Works. Frame writes are serialized per socket by a mutex.
+// GPU tensors are staged through host memory by the Python wrapper (v1).
+#pragma once
+
+#include <atomic>
+#include <memory>
+#include <mutex>
+#include <thread>
+#include <unordered_map>
+#include <vector>
+
+#include "common.h"
+#include "cpu_comm.h"
+#include "tcp.h"
+
+namespace tma {
+
+class PsContext {
+ public:
+  PsContext(int rank, const std::string& bind_addr);
+  ~PsContext();
+
+  int listen_port() const { return mesh_.listen_port(); }
+  // Connect the PS mesh and start the event loop.
+  void start(int rank, const std::vector<std::string>& addrs);
+  void stop();
+
+  // Register tensor `tid` of `count` elements; `init` points at the full
+  // initial value (every rank passes the same count; each rank stores its
+  // own shard from its slice of init).
+  void register_tensor(int tid, size_t count, DType d, const void* init);
+  void free_tensor(int tid);
+
+  // Push `data` (full tensor) shard-wise to the owners with `rule`
+  // (0 = copy, 1 = add). Completes when every owner has ACKed.
+  std::shared_ptr<Work> send(int tid, const void* data, int rule);
+  // Gather the current shard values into `data` (full tensor).
+  std::shared_ptr<Work> receive(int tid, void* data);
+
+  // Shard layout for tensor tid: offset/count (elements) of rank r's shard.
+  void shard_of(int tid, int r, size_t& off, size_t& cnt);
+
+ private:
+  struct Reg {
+    size_t count = 0;
+    DType dtype = DType::F32;
+    std::vector<char> shard;  // this rank's shard storage
+    std::vector<size_t> off, cnt;
+    std::mutex m;
+  };
+
+  struct Pending {
+    std::shared_ptr<Work> work;
+    std::atomic<int> remaining{0};
+    void* dest = nullptr;  // for receive: full-tensor base pointer
+    int tid = 0;
+  };
+
+  void loop();
+  void handle_frame(int peer);
+  int mesh_fd(int peer) { return mesh_.fd(peer); }
+  void write_frame(int peer, const void* hdr, size_t hdr_n, const void* payload,
+                   size_t payload_n);
+  Reg& reg(int tid);
+
+  int rank_ = 0;
+  int world_ = 1;
+  TcpMesh mesh_;
+  std::thread loop_thread_;
+  std::atomic<bool> running_{false};
+  int wake_pipe_[2] = {-1, -1};
+
+  std::mutex regs_m_;
+  std::unordered_map<int, std::unique_ptr<Reg>> regs_;
+
+  std::mutex pend_m_;
+  std::unordered_map<uint64_t, std::shared_ptr<Pending>> pending_;
+  std::atomic<uint64_t> next_seq_{1};
+
+  std::vector<std::unique_ptr<std::mutex>> sock_m_;  // per-peer write locks
+};
+
+}  // namespace tma
