@@ -1,0 +1,148 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ResNet-50 DP-SGD with async bucketed gradient
+allreduce overlapped with backward (BASELINE.json configs #2/#3).
+
+Run (single GPU):   python bench.py --gpus 1 --steps 20 --warmup 5
+Run (N GPUs):       python -m torch.distributed.run --nnodes=1 \
+                      --nproc-per-node N --master-addr 127.0.0.1 bench.py \
+                      --gpus N --steps 20 --warmup 5
+
+Prints ONE JSON line from rank 0:
+  value = whole-job samples/sec (aggregate over all GPUs, weak scaling);
+  config also reports the measured 256 MB bf16 allreduce bus bandwidth
+  (the BASELINE headline collective metric) when world_size > 1.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--local-batch", type=int, default=256)
+    p.add_argument("--bucket-mb", type=int, default=25)
+    return p.parse_args()
+
+
+def measure_allreduce_busbw(tm, device, size_bytes=256 << 20, iters=10):
+    """NCCL-convention bus bandwidth of a single-tensor bf16 ring allreduce."""
+    W = tm.size()
+    if W < 2 or device.type != "cuda":
+        return None
+    n = size_bytes // 2
+    t = torch.ones(n, dtype=torch.bfloat16, device=device)
+    for _ in range(3):
+        tm.allreduce(t)
+    torch.cuda.synchronize()
+    tm.barrier()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        tm.allreduce(t)
+    torch.cuda.synchronize()
+    el = (time.perf_counter() - t0) / iters
+    tm.barrier()
+    busbw = (2 * (W - 1) / W) * size_bytes / el / 1e9
+    return round(busbw, 2)
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    use_gpu = torch.cuda.is_available()
+
+    import torchmpi_amd as tm
+    from torchmpi_amd.models import resnet50
+    from torchmpi_amd.nn import GradientSynchronizer, synchronize_parameters
+
+    tm.start(use_gpu=use_gpu)
+    rank = tm.rank()
+    W = tm.size()
+    if use_gpu:
+        device = torch.device("cuda", torch.cuda.current_device())
+        dtype = torch.bfloat16
+        local_batch = args.local_batch
+    else:  # CPU smoke path (dev sandbox only)
+        device = torch.device("cpu")
+        dtype = torch.float32
+        local_batch = 8
+
+    model = resnet50().to(device=device, dtype=dtype)
+    synchronize_parameters(model)
+    sync = GradientSynchronizer(model, bucket_mb=args.bucket_mb, average=True)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+
+    img = 224 if use_gpu else 64
+    x = torch.randn(local_batch, 3, img, img, device=device, dtype=dtype)
+    y = torch.randint(0, 1000, (local_batch,), device=device)
+
+    def step():
+        opt.zero_grad(set_to_none=False)
+        out = model(x)
+        loss = torch.nn.functional.cross_entropy(out.float(), y)
+        loss.backward()
+        sync.finish()
+        opt.step()
+        return loss
+
+    for _ in range(args.warmup):
+        step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    tm.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    tm.barrier()
+
+    # Job time = max over ranks.
+    el_t = torch.tensor([elapsed], dtype=torch.float64)
+    all_el = tm.allgather(el_t)
+    elapsed_max = float(all_el.max())
+
+    busbw = measure_allreduce_busbw(tm, device)
+
+    if rank == 0:
+        samples_per_sec = W * local_batch * args.steps / elapsed_max
+        cfg = {
+            "model": "resnet50",
+            "global_batch": W * local_batch,
+            "local_batch": local_batch,
+            "image_size": img,
+            "parallelism": f"dp{W}",
+            "overlap": "bucketed async allreduce over backward",
+        }
+        if busbw is not None:
+            cfg["allreduce_busbw_256MB_bf16_GBps"] = busbw
+        print(json.dumps({
+            "metric": "DP-SGD samples/sec, ResNet-50 (+ allreduce bus GB/s 256MB bf16 in config)",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": W,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed_max / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": cfg,
+        }))
+
+    tm.stop()
+
+
+if __name__ == "__main__":
+    main()

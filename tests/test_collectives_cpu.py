@@ -1,0 +1,159 @@
+"""Collectives sweep over the TCP transport (reference test/collectives.lua:
+verb x dtype x size x sync/async with analytic expected values — SURVEY.md
+section 4)."""
+import pytest
+import torch
+
+from .util import run_distributed
+
+SIZES = [1, 7, 64, 1000, 65536, 1 << 20]
+DTYPES = [torch.float32, torch.float64, torch.int32, torch.int64,
+          torch.bfloat16, torch.float16, torch.uint8]
+
+
+def _tol(dtype):
+    return {torch.bfloat16: 0.05, torch.float16: 1e-2}.get(dtype, 1e-6)
+
+
+def _allreduce_body(tm):
+    W, r = tm.size(), tm.rank()
+    for dtype in DTYPES:
+        for n in SIZES:
+            if dtype in (torch.bfloat16, torch.float16) and n > 65536:
+                continue
+            t = torch.full((n,), float(r + 1)).to(dtype)
+            tm.allreduce(t)
+            expect = torch.full((n,), float(W * (W + 1) / 2)).to(dtype)
+            assert torch.allclose(t.float(), expect.float(),
+                                  rtol=_tol(dtype)), (dtype, n, t[:4])
+    # max / min / prod
+    t = torch.full((100,), float(r + 1))
+    tm.allreduce(t, op="max")
+    assert torch.equal(t, torch.full((100,), float(W)))
+    t = torch.full((100,), float(r + 1))
+    tm.allreduce(t, op="min")
+    assert torch.equal(t, torch.full((100,), 1.0))
+    t = torch.full((100,), float(r + 1))
+    tm.allreduce(t, op="prod")
+    import math
+    assert torch.equal(t, torch.full((100,), float(math.factorial(W))))
+
+
+def test_allreduce_w2():
+    run_distributed(_allreduce_body, world=2)
+
+
+def test_allreduce_w4():
+    run_distributed(_allreduce_body, world=4)
+
+
+def _broadcast_body(tm):
+    W, r = tm.size(), tm.rank()
+    for root in range(W):
+        for n in [1, 1000, 1 << 20]:
+            t = (torch.arange(n, dtype=torch.float32) + root if r == root
+                 else torch.zeros(n))
+            tm.broadcast(t, root=root)
+            assert torch.equal(t, torch.arange(n, dtype=torch.float32) + root)
+
+
+def test_broadcast():
+    run_distributed(_broadcast_body, world=4)
+
+
+def _reduce_body(tm):
+    W, r = tm.size(), tm.rank()
+    for root in range(W):
+        t = torch.full((1000,), float(r + 1))
+        tm.reduce(t, root=root)
+        if r == root:
+            assert torch.equal(t, torch.full((1000,), float(W * (W + 1) / 2)))
+        else:
+            # non-root input preserved (MPI_Reduce semantics)
+            assert torch.equal(t, torch.full((1000,), float(r + 1)))
+
+
+def test_reduce():
+    run_distributed(_reduce_body, world=4)
+
+
+def _allgather_body(tm):
+    W, r = tm.size(), tm.rank()
+    for n in [1, 513, 100000]:
+        out = tm.allgather(torch.full((n,), float(r)))
+        assert out.shape == (W, n)
+        for k in range(W):
+            assert torch.equal(out[k], torch.full((n,), float(k)))
+
+
+def test_allgather():
+    run_distributed(_allgather_body, world=4)
+
+
+def _reduce_scatter_body(tm):
+    W, r = tm.size(), tm.rank()
+    n = 1000
+    t = torch.arange(n, dtype=torch.float32) * (r + 1)
+    base, rem = divmod(n, W)
+    cnt = base + (1 if r < rem else 0)
+    off = r * base + min(r, rem)
+    out = torch.zeros(cnt)
+    tm.reduce_scatter(t, out)
+    s = W * (W + 1) / 2
+    expect = torch.arange(off, off + cnt, dtype=torch.float32) * s
+    assert torch.allclose(out, expect), (out[:4], expect[:4])
+
+
+def test_reduce_scatter():
+    run_distributed(_reduce_scatter_body, world=4)
+    run_distributed(_reduce_scatter_body, world=3)
+
+
+def _sendrecv_body(tm):
+    W, r = tm.size(), tm.rank()
+    s = torch.full((100,), float(r))
+    rv = torch.zeros(100)
+    tm.sendreceive(s, (r + 1) % W, rv, (r - 1) % W)
+    assert torch.equal(rv, torch.full((100,), float((r - 1) % W)))
+    # plain send/recv pair
+    if r == 0:
+        tm.send(torch.arange(10.0), dst=1)
+    elif r == 1:
+        t = torch.zeros(10)
+        tm.recv(t, src=0)
+        assert torch.equal(t, torch.arange(10.0))
+
+
+def test_sendrecv():
+    run_distributed(_sendrecv_body, world=4)
+
+
+def _alltoall_body(tm):
+    W, r = tm.size(), tm.rank()
+    t = torch.arange(W, dtype=torch.float32) + r * 10
+    out = tm.alltoall(t)
+    expect = torch.tensor([r + k * 10.0 for k in range(W)])
+    assert torch.equal(out, expect), (out, expect)
+
+
+def test_alltoall():
+    run_distributed(_alltoall_body, world=4)
+
+
+def _noncontig_rejected_body(tm):
+    t = torch.zeros(4, 4).t()
+    with pytest.raises(RuntimeError):
+        tm.allreduce(t)
+
+
+def test_noncontiguous_rejected():
+    run_distributed(_noncontig_rejected_body, world=2)
+
+
+def _barrier_body(tm):
+    for _ in range(20):
+        tm.barrier()
+
+
+def test_barrier():
+    run_distributed(_barrier_body, world=4)
