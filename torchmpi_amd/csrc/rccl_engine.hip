@@ -89,6 +89,22 @@ void split_aligned(size_t count, int parts, size_t align, std::vector<size_t>& o
 
 }  // namespace
 
+// The single source of truth for the ring schedules (used by the engine AND
+// exported to Python for simulation tests). phase 0 = reduce-scatter,
+// phase 1 = allgather. Position p at step `step` sends segment send_seg to
+// its ring successor and receives recv_seg from its predecessor; after the
+// RS phase rank g owns segment g for any ring permutation.
+inline void ring_step_segs(const std::vector<int>& perm, int W, int p, int step,
+                           int phase, int& send_seg, int& recv_seg) {
+  if (phase == 0) {
+    send_seg = perm[(p - step - 1 + 2 * W) % W];
+    recv_seg = perm[(p - step - 2 + 2 * W) % W];
+  } else {
+    send_seg = perm[(p - step + W) % W];
+    recv_seg = perm[(p - step - 1 + 2 * W) % W];
+  }
+}
+
 // Completion handle: one hipEvent per comm stream used by the op.
 struct GpuWork {
   std::vector<hipEvent_t> events;
@@ -486,8 +502,8 @@ class RcclEngine {
     char* sc = static_cast<char*>(scratch.data_ptr());
     const int p = pos_[r];
     for (int step = 0; step < W - 1; ++step) {
-      int send_seg = perm[(p - step - 1 + 2 * W) % W];
-      int recv_seg = perm[(p - step - 2 + 2 * W) % W];
+      int send_seg, recv_seg;
+      ring_step_segs(perm, W, p, step, /*phase=*/0, send_seg, recv_seg);
       char* rbuf = sc + (step % 2) * cnt[0] * es;
       TMA_NCCL_CHECK(ncclGroupStart());
       if (cnt[send_seg] > 0)
@@ -517,8 +533,8 @@ class RcclEngine {
     split_segments(count, W, off, cnt);
     const int p = pos_[r];
     for (int step = 0; step < W - 1; ++step) {
-      int send_seg = perm[(p - step + W) % W];
-      int recv_seg = perm[(p - step - 1 + 2 * W) % W];
+      int send_seg, recv_seg;
+      ring_step_segs(perm, W, p, step, /*phase=*/1, send_seg, recv_seg);
       TMA_NCCL_CHECK(ncclGroupStart());
       if (cnt[send_seg] > 0)
         TMA_NCCL_CHECK(ncclSend(base + off[send_seg] * es, cnt[send_seg] * es,
@@ -609,6 +625,28 @@ std::string py_get_unique_id() {
   return std::string(reinterpret_cast<char*>(&id), sizeof(id));
 }
 
+// The exact ring schedules the engine executes, exposed for simulation tests
+// (the 8-GPU path cannot run on the 1-GPU CI box; tests/test_ring_schedule.py
+// simulates these step lists and checks the allreduce/reduce-scatter
+// invariants for every world size x ring permutation).
+// phase 0 = reduce-scatter, phase 1 = allgather.
+// Returns steps x world x 2: [step][position] = (send_seg, recv_seg).
+std::vector<std::vector<std::vector<int>>> py_ring_schedule(
+    int world, std::vector<int> perm, int phase) {
+  const int W = world;
+  std::vector<std::vector<std::vector<int>>> out;
+  for (int step = 0; step < W - 1; ++step) {
+    std::vector<std::vector<int>> row;
+    for (int p = 0; p < W; ++p) {
+      int send_seg, recv_seg;
+      ring_step_segs(perm, W, p, step, phase, send_seg, recv_seg);
+      row.push_back({send_seg, recv_seg});
+    }
+    out.push_back(row);
+  }
+  return out;
+}
+
 }  // namespace
 
 void register_gpu(pybind11::module_& m) {
@@ -648,6 +686,8 @@ void register_gpu(pybind11::module_& m) {
       .def("recv", &RcclEngine::recv, py::call_guard<py::gil_scoped_release>());
 
   m.def("nccl_unique_id", [] { return pybind11::bytes(py_get_unique_id()); });
+  m.def("ring_schedule", &py_ring_schedule,
+        "engine ring schedule for simulation tests");
   m.def("reduce_into", &py_reduce_into, "dst = dst (op) src on current stream");
   m.def("scale_", &py_scale);
   m.def("build_pack_meta", &py_build_pack_meta);
