@@ -73,12 +73,18 @@ def main():
         local_batch = 8
 
     model = resnet50().to(device=device, dtype=dtype)
+    if use_gpu:
+        # NHWC: MIOpen's fast igemm conv kernels (NCHW bf16 falls back to
+        # naive_conv — 81% of step time, measured in profiles/).
+        model = model.to(memory_format=torch.channels_last)
     synchronize_parameters(model)
     sync = GradientSynchronizer(model, bucket_mb=args.bucket_mb, average=True)
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
 
     img = 224 if use_gpu else 64
     x = torch.randn(local_batch, 3, img, img, device=device, dtype=dtype)
+    if use_gpu:
+        x = x.contiguous(memory_format=torch.channels_last)
     y = torch.randint(0, 1000, (local_batch,), device=device)
 
     def step():

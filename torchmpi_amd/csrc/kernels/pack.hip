@@ -133,9 +133,11 @@ __global__ __launch_bounds__(kBlock) void unpack_kernel(
   }
 }
 
-inline int grid_for(unsigned long long work) {
+// Same measured heuristic as reduce.hip: small grid for big streams.
+inline int grid_for(unsigned long long work, unsigned long long bytes_streamed) {
+  unsigned long long cap = bytes_streamed >= (64u << 20) ? 512 : kMaxBlocks;
   unsigned long long blocks = (work + kBlock - 1) / kBlock;
-  if (blocks > kMaxBlocks) blocks = kMaxBlocks;
+  if (blocks > cap) blocks = cap;
   if (blocks == 0) blocks = 1;
   return static_cast<int>(blocks);
 }
@@ -147,7 +149,8 @@ void launch_pack(void* bucket, DType bucket_dtype, const void* metas_dev, int nt
                  size_t total_elems, DType tensor_dtype, hipStream_t stream) {
   TMA_CHECK(total_elems % (16 / dtype_size(bucket_dtype)) == 0,
             "bucket size must be 16B-aligned");
-  int grid = grid_for(total_elems / (16 / dtype_size(bucket_dtype)));
+  int grid = grid_for(total_elems / (16 / dtype_size(bucket_dtype)),
+                      2 * total_elems * dtype_size(bucket_dtype));
   auto metas = static_cast<const PackMeta*>(metas_dev);
 #define TMA_PACK(TB, TT)                                                              \
   hipLaunchKernelGGL((pack_kernel<TB, TT>), dim3(grid), dim3(kBlock), 0, stream,      \
@@ -168,7 +171,8 @@ void launch_pack(void* bucket, DType bucket_dtype, const void* metas_dev, int nt
 void launch_unpack(const void* bucket, DType bucket_dtype, const void* metas_dev,
                    int ntensors, size_t total_elems, DType tensor_dtype, double alpha,
                    hipStream_t stream) {
-  int grid = grid_for(total_elems / (16 / dtype_size(bucket_dtype)));
+  int grid = grid_for(total_elems / (16 / dtype_size(bucket_dtype)),
+                      2 * total_elems * dtype_size(bucket_dtype));
   auto metas = static_cast<const PackMeta*>(metas_dev);
   float a = static_cast<float>(alpha);
 #define TMA_UNPACK(TB, TT)                                                            \

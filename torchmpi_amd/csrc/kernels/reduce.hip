@@ -135,9 +135,13 @@ __global__ __launch_bounds__(kBlock) void scale_vec_kernel(T* __restrict__ data,
         from_acc<T>(to_acc(data[tail]) * static_cast<typename AccOf<T>::type>(alpha));
 }
 
-inline int grid_for(size_t work_items) {
+// Measured on MI355X (benchmarks/probe_reduce.hip, profiles/): at >= 64 MB
+// streamed, 512 blocks (2/CU) beats 2048 by ~8% (less DVFS/L2 pressure);
+// below that the larger grid wins slightly.
+inline int grid_for(size_t work_items, size_t bytes_streamed) {
+  size_t cap = bytes_streamed >= (64u << 20) ? 512 : kMaxBlocks;
   size_t blocks = (work_items + kBlock - 1) / kBlock;
-  if (blocks > kMaxBlocks) blocks = kMaxBlocks;
+  if (blocks > cap) blocks = cap;
   if (blocks == 0) blocks = 1;
   return static_cast<int>(blocks);
 }
@@ -148,7 +152,7 @@ void launch_reduce_t(void* dst, const void* src, size_t n, RedOp op,
   bool aligned = ((reinterpret_cast<uintptr_t>(dst) | reinterpret_cast<uintptr_t>(src)) &
                   15) == 0;
   constexpr int VEC = 16 / sizeof(T);
-  int grid = grid_for(aligned ? (n / VEC + 1) : n);
+  int grid = grid_for(aligned ? (n / VEC + 1) : n, 3 * n * sizeof(T));
   auto d = static_cast<T*>(dst);
   auto s = static_cast<const T*>(src);
 #define TMA_DISPATCH_OP(OPI)                                                          \
@@ -186,7 +190,7 @@ void launch_reduce(void* dst, const void* src, size_t count, DType d, RedOp op,
 }
 
 void launch_scale(void* data, size_t count, DType d, double alpha, hipStream_t stream) {
-  int grid = grid_for(count / 4 + 1);
+  int grid = grid_for(count / 4 + 1, 2 * count * dtype_size(d));
   switch (d) {
     case DType::F32:
       hipLaunchKernelGGL((scale_vec_kernel<float>), dim3(grid), dim3(kBlock), 0, stream,

@@ -593,7 +593,11 @@ at::Tensor py_build_pack_meta(std::vector<at::Tensor> tensors,
                         at::TensorOptions().dtype(at::kByte));
   auto* metas = reinterpret_cast<HostPackMeta*>(host.data_ptr());
   for (size_t i = 0; i < tensors.size(); ++i) {
-    check_gpu_tensor(tensors[i]);
+    // Dense non-overlapping is enough (channels_last grads etc.): pack copies
+    // in raw storage order, identical on every rank.
+    TMA_CHECK(tensors[i].is_cuda(), "pack: expected GPU tensors");
+    TMA_CHECK(tensors[i].is_non_overlapping_and_dense(),
+              "pack: tensors must be dense");
     metas[i].ptr = tensors[i].data_ptr();
     metas[i].off = static_cast<unsigned long long>(offsets[i]);
     metas[i].n = static_cast<unsigned long long>(tensors[i].numel());
