@@ -72,25 +72,30 @@ def main():
         dtype = torch.float32
         local_batch = 8
 
-    model = resnet50().to(device=device, dtype=dtype)
+    # fp32 master weights + bf16 autocast compute; gradients communicated as
+    # bf16 via the cast-fused bucket pack kernels (BASELINE config #4's
+    # flat-view + compression path). NHWC for MIOpen's fast igemm convs
+    # (NCHW bf16 falls back to naive_conv — 81% of step time, see profiles/).
+    model = resnet50().to(device=device)
     if use_gpu:
-        # NHWC: MIOpen's fast igemm conv kernels (NCHW bf16 falls back to
-        # naive_conv — 81% of step time, measured in profiles/).
         model = model.to(memory_format=torch.channels_last)
+        torch.backends.cudnn.benchmark = True  # MIOpen algo search in warmup
     synchronize_parameters(model)
-    sync = GradientSynchronizer(model, bucket_mb=args.bucket_mb, average=True)
+    sync = GradientSynchronizer(model, bucket_mb=args.bucket_mb, average=True,
+                                comm_dtype=torch.bfloat16 if use_gpu else None)
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
 
     img = 224 if use_gpu else 64
-    x = torch.randn(local_batch, 3, img, img, device=device, dtype=dtype)
+    x = torch.randn(local_batch, 3, img, img, device=device, dtype=torch.float32)
     if use_gpu:
         x = x.contiguous(memory_format=torch.channels_last)
     y = torch.randint(0, 1000, (local_batch,), device=device)
 
     def step():
         opt.zero_grad(set_to_none=False)
-        out = model(x)
-        loss = torch.nn.functional.cross_entropy(out.float(), y)
+        with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_gpu):
+            out = model(x)
+            loss = torch.nn.functional.cross_entropy(out.float(), y)
         loss.backward()
         sync.finish()
         opt.step()
