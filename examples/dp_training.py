@@ -1,0 +1,86 @@
+#!/usr/bin/env python3
+"""Synchronous data-parallel SGD with overlapped gradient allreduce
+(reference: examples/mnist/mnist_allreduce.lua, SURVEY.md stack 3.5).
+
+Run:  python examples/dp_training.py --world 2          (spawn, CPU)
+      torchrun --standalone --local-addr 127.0.0.1 --nproc-per-node 8 \
+          examples/dp_training.py                        (one rank per GPU)
+
+Synthetic data (no network access for datasets); each rank trains on its
+own shard (shard = sample_index % world, like the reference examples).
+"""
+import argparse
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def make_model():
+    torch.manual_seed(1234)
+    return torch.nn.Sequential(
+        torch.nn.Flatten(),
+        torch.nn.Linear(28 * 28, 256), torch.nn.ReLU(),
+        torch.nn.Linear(256, 128), torch.nn.ReLU(),
+        torch.nn.Linear(128, 10),
+    )
+
+
+def main_worker():
+    import torchmpi_amd as tm
+    from torchmpi_amd.nn import GradientSynchronizer, synchronize_parameters
+
+    tm.start()
+    rank, W = tm.rank(), tm.size()
+    dev = (torch.device("cuda", torch.cuda.current_device())
+           if torch.cuda.is_available() else torch.device("cpu"))
+
+    model = make_model().to(dev)
+    synchronize_parameters(model)          # replicas agree at t=0
+    sync = GradientSynchronizer(model, average=True)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
+
+    # Synthetic "dataset": 4096 samples, rank-sharded sample_index % W.
+    torch.manual_seed(999)
+    data = torch.randn(4096, 1, 28, 28)
+    labels = (data.view(4096, -1).sum(-1) * 10).long().abs() % 10
+    my_idx = torch.arange(rank, 4096, W)
+
+    for epoch in range(3):
+        perm = torch.randperm(len(my_idx))
+        total_loss, n = 0.0, 0
+        for i in range(0, len(my_idx), 64):
+            batch = my_idx[perm[i:i + 64]]
+            x, y = data[batch].to(dev), labels[batch].to(dev)
+            opt.zero_grad(set_to_none=False)
+            loss = torch.nn.functional.cross_entropy(model(x), y)
+            loss.backward()           # bucket allreduces fire from hooks
+            sync.finish()             # wait + unpack (grads now averaged)
+            opt.step()
+            total_loss += float(loss) * len(batch)
+            n += len(batch)
+        if rank == 0:
+            print(f"epoch {epoch}: loss {total_loss / n:.4f}")
+    tm.barrier()
+    tm.stop()
+
+
+def _spawn(rank, world, port):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      LOCAL_RANK=str(rank))
+    main_worker()
+
+
+if __name__ == "__main__":
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--world", type=int, default=0,
+                    help="spawn N local processes (else use torchrun env)")
+    args = ap.parse_args()
+    if args.world > 0:
+        import torch.multiprocessing as mp
+        mp.spawn(_spawn, (args.world, 29610), nprocs=args.world, join=True)
+    else:
+        main_worker()
