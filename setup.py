@@ -19,6 +19,7 @@ sources = [
     os.path.join(CSRC, "cpu_comm.cpp"),
     os.path.join(CSRC, "ps.cpp"),
     os.path.join(CSRC, "rccl_engine.hip"),
+    os.path.join(CSRC, "ipc_ps.hip"),
     os.path.join(CSRC, "kernels", "reduce.hip"),
     os.path.join(CSRC, "kernels", "pack.hip"),
 ]

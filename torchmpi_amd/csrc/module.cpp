@@ -18,7 +18,8 @@ namespace py = pybind11;
 
 namespace tma {
 
-void register_gpu(py::module_& m);  // defined in rccl_engine.hip
+void register_gpu(py::module_& m);     // defined in rccl_engine.hip
+void register_ipc_ps(py::module_& m);  // defined in ipc_ps.hip
 
 namespace {
 
@@ -186,8 +187,9 @@ PYBIND11_MODULE(_C, m) {
              return p.receive(tid, t.data_ptr());
            });
 
-  // GPU engine + gfx950 kernels.
+  // GPU engine + gfx950 kernels + xGMI IPC parameter server.
   register_gpu(m);
+  register_ipc_ps(m);
 }
 
 }  // namespace tma

@@ -1,13 +1,19 @@
 """Sharded parameter server (reference: torchmpi/parameterserver.lua +
 lib/parameterserver.cpp, SURVEY.md sections 2.1/3.4).
 
-Each registered tensor is sharded round-robin-contiguously across ranks;
-clients push updates with 'copy' or 'add' rules and pull values
-asynchronously. Used to implement Downpour-SGD and EASGD
-(examples/downpour_mnist.py, examples/easgd.py).
+Each registered tensor is sharded contiguously across ranks; clients push
+updates with 'copy' or 'add' rules and pull values asynchronously. Used to
+implement Downpour-SGD and EASGD (examples/downpour.py, examples/easgd.py).
 
-GPU tensors are staged through host memory (the PS is the async-SGD control
-path; the bandwidth-critical path is the collective engine).
+Two transports, picked per tensor:
+  - GPU tensors: ONE-SIDED over xGMI — shards live in fine-grained device
+    memory mapped into every client via hipIpc; 'copy'/receive are remote
+    copy kernels, 'add' is a system-scope atomicAdd kernel (csrc/ipc_ps.hip).
+    No server thread exists; the fabric is the server. The reference needed
+    an MPI server loop for this — on a fully-connected MI355X
+    node it reduces to peer mappings.
+  - CPU tensors: the TCP-mesh PS with a per-rank event-loop thread
+    (csrc/ps.cpp), matching the reference's design.
 """
 from typing import Dict, Optional
 
@@ -21,19 +27,27 @@ _RULES = {"copy": 0, "add": 1}
 
 
 class PsHandle:
-    def __init__(self, work, post=None):
+    def __init__(self, work=None, event=None, post=None):
         self._work = work
+        self._event = event
         self._post = post
 
     def wait(self):
-        self._work.wait()
+        if self._work is not None:
+            self._work.wait()
+        if self._event is not None:
+            self._event.synchronize()
         if self._post is not None:
             self._post()
             self._post = None
         return self
 
     def ready(self):
-        return self._work.ready()
+        if self._work is not None:
+            return self._work.ready()
+        if self._event is not None:
+            return self._event.query()
+        return True
 
 
 class ParameterServer:
@@ -45,26 +59,41 @@ class ParameterServer:
         from . import communicator
         self.comm = comm or communicator()
         ParameterServer._counter += 1
-        rdv: Rendezvous = self.comm.rdv.scoped(f"ps{ParameterServer._counter}")
+        self.rdv: Rendezvous = self.comm.rdv.scoped(f"ps{ParameterServer._counter}")
         bind = env_str("TORCHMPI_BIND_ADDR", "127.0.0.1")
         self._ctx = _C.PsContext(self.comm.rank(), bind)
         my_addr = f"{bind}:{self._ctx.listen_port()}".encode()
-        addrs = rdv.exchange("psaddr", my_addr, ranks=self.comm.global_ranks)
+        addrs = self.rdv.exchange("psaddr", my_addr, ranks=self.comm.global_ranks)
         self._ctx.start(self.comm.rank(), [a.decode() for a in addrs])
+        self._gpu = None  # lazy: created on first GPU registration
         self._tids: Dict[int, int] = {}
-        self._staging: Dict[int, torch.Tensor] = {}
+        self._kind: Dict[int, str] = {}
         self._next_tid = 0
 
+    def _gpu_ctx(self):
+        if self._gpu is None:
+            dev = (self.comm.device
+                   if self.comm.device is not None
+                   else torch.device("cuda", torch.cuda.current_device()))
+            self._gpu = _C.IpcPs(self.comm.rank(), self.comm.size(), dev.index)
+        return self._gpu
+
     def register(self, tensor: torch.Tensor) -> int:
-        """Collective: shard `tensor`'s current value across ranks.
-        Returns the tensor id."""
+        """Collective: shard `tensor`'s current value across ranks."""
         tid = self._next_tid
         self._next_tid += 1
-        host = tensor.detach().reshape(-1).cpu().contiguous()
-        self._ctx.register_tensor(tid, host)
-        self._tids[id(tensor)] = tid
+        flat = tensor.detach().reshape(-1).contiguous()
         if tensor.is_cuda:
-            self._staging[tid] = torch.empty_like(host, pin_memory=True)
+            ctx = self._gpu_ctx()
+            handle = ctx.register_local(tid, flat)
+            handles = self.rdv.exchange(f"ipc{tid}", handle,
+                                        ranks=self.comm.global_ranks)
+            ctx.open_peers(tid, [bytes(h) for h in handles])
+            self._kind[tid] = "gpu"
+        else:
+            self._ctx.register_tensor(tid, flat)
+            self._kind[tid] = "cpu"
+        self._tids[id(tensor)] = tid
         return tid
 
     def _tid_of(self, tensor: torch.Tensor) -> int:
@@ -78,38 +107,46 @@ class ParameterServer:
         """Push `data` (default: the registered tensor's value) to the shard
         owners with `rule` in {'copy','add'}."""
         tid = self._tid_of(tensor)
-        src = tensor if data is None else data
-        flat = src.detach().reshape(-1)
-        if flat.is_cuda:
-            stage = self._staging[tid]
-            stage.copy_(flat, non_blocking=False)
-            flat = stage
-        else:
-            flat = flat.contiguous()
-        return PsHandle(self._ctx.send(tid, flat, _RULES[rule]))
+        src = (tensor if data is None else data).detach().reshape(-1).contiguous()
+        if self._kind[tid] == "gpu":
+            self._gpu.send(tid, src, _RULES[rule])
+            ev = torch.cuda.Event()
+            ev.record()
+            return PsHandle(event=ev)
+        return PsHandle(work=self._ctx.send(tid, src, _RULES[rule]))
 
     def receive(self, tensor: torch.Tensor,
                 out: Optional[torch.Tensor] = None) -> PsHandle:
         """Pull the current sharded value of registered `tensor` into `out`
         (default: the tensor itself), async."""
         tid = self._tid_of(tensor)
-        flat = (tensor if out is None else out).detach().reshape(-1)
-        if flat.is_cuda:
-            stage = self._staging[tid]
-            w = self._ctx.receive(tid, stage)
+        dst = (tensor if out is None else out).detach().reshape(-1)
+        if self._kind[tid] == "gpu":
+            assert dst.is_contiguous(), "ps receive: need contiguous destination"
+            self._gpu.receive(tid, dst)
+            ev = torch.cuda.Event()
+            ev.record()
+            return PsHandle(event=ev)
+        flat = dst.contiguous()
+        if flat.data_ptr() != dst.data_ptr():
+            w = self._ctx.receive(tid, flat)
 
             def post():
-                flat.copy_(stage)
-            return PsHandle(w, post)
-        return PsHandle(self._ctx.receive(tid, flat))
+                dst.copy_(flat)
+            return PsHandle(work=w, post=post)
+        return PsHandle(work=self._ctx.receive(tid, flat))
 
     prefetch = receive  # reference: prefetch == async receive
 
     def free(self, tensor: torch.Tensor) -> None:
         tid = self._tids.pop(id(tensor), None)
         if tid is not None:
-            self._ctx.free_tensor(tid)
-            self._staging.pop(tid, None)
+            if self._kind.get(tid) == "gpu":
+                self._gpu.free_tensor(tid)
+            else:
+                self._ctx.free_tensor(tid)
+            self._kind.pop(tid, None)
 
     def shutdown(self):
         self._ctx.stop()
+        self._gpu = None
