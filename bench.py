@@ -30,6 +30,31 @@ def parse_args():
     return p.parse_args()
 
 
+def verify_allreduce(tm, device):
+    """Analytic check of the engine's composed allreduce on THIS topology:
+    iteration 0 runs the eager multi-ring path (and captures the hipGraph),
+    iterations 1-2 replay the graph. Eager wrong => hard failure (no silent
+    bad numbers); replay wrong => disable graphs, keep eager."""
+    W = tm.size()
+    if W < 2 or device.type != "cuda":
+        return "n/a"
+    n = 1 << 20  # 4 MB fp32: multi-ring + graph eligible
+    t = torch.empty(n, device=device)
+    expect = float(W * (W + 1) / 2)
+    for it in range(3):
+        t.fill_(float(tm.rank() + 1))
+        tm.allreduce(t)
+        torch.cuda.synchronize()
+        ok = bool(torch.allclose(t, torch.full_like(t, expect)))
+        if not ok and it == 0:
+            raise RuntimeError("eager multi-ring allreduce produced wrong "
+                               "values — refusing to benchmark")
+        if not ok:
+            tm.communicator().engine.set_graphs_enabled(False)
+            return "graphs-disabled(replay-mismatch)"
+    return "ok"
+
+
 def measure_allreduce_busbw(tm, device, size_bytes=256 << 20, iters=10):
     """NCCL-convention bus bandwidth of a single-tensor bf16 ring allreduce."""
     W = tm.size()
@@ -85,6 +110,8 @@ def main():
                                 comm_dtype=torch.bfloat16 if use_gpu else None)
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
 
+    graph_status = verify_allreduce(tm, device)
+
     img = 224 if use_gpu else 64
     x = torch.randn(local_batch, 3, img, img, device=device, dtype=torch.float32)
     if use_gpu:
@@ -136,6 +163,8 @@ def main():
         }
         if busbw is not None:
             cfg["allreduce_busbw_256MB_bf16_GBps"] = busbw
+        if graph_status != "n/a":
+            cfg["allreduce_selfcheck"] = graph_status
         print(json.dumps({
             "metric": "DP-SGD samples/sec, ResNet-50 (+ allreduce bus GB/s 256MB bf16 in config)",
             "value": round(samples_per_sec, 2),

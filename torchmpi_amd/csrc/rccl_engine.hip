@@ -28,7 +28,9 @@
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
+#include <cstdio>
 #include <memory>
+#include <unordered_map>
 #include <vector>
 
 #include "common.h"
@@ -172,15 +174,25 @@ class RcclEngine {
     chunk_bytes_ = cb ? static_cast<size_t>(std::atoll(cb)) : (4u << 20);
     const char* mr = std::getenv("TORCHMPI_MULTIRING_MIN_BYTES");
     multiring_min_bytes_ = mr ? static_cast<size_t>(std::atoll(mr)) : (1u << 20);
+    const char* ge = std::getenv("TORCHMPI_GRAPH");
+    graphs_enabled_ = !(ge && std::string(ge) == "0");
+    const char* gm = std::getenv("TORCHMPI_GRAPH_MIN_BYTES");
+    if (gm) graph_min_bytes_ = static_cast<size_t>(std::atoll(gm));
   }
 
   ~RcclEngine() {
+    for (auto& kv : graphs_) {
+      if (kv.second.exec) (void)hipGraphExecDestroy(kv.second.exec);
+      for (auto ev : kv.second.events) (void)hipEventDestroy(ev);
+    }
     for (auto c : comms_) (void)ncclCommDestroy(c);
   }
 
   int rank() const { return rank_; }
   int size() const { return world_; }
   int num_rings() const { return static_cast<int>(comms_.size()); }
+  bool graphs_enabled() const { return graphs_enabled_; }
+  void set_graphs_enabled(bool v) { graphs_enabled_ = v; }
 
   // ---- verbs -------------------------------------------------------------
 
@@ -207,6 +219,27 @@ class RcclEngine {
     std::vector<size_t> soff, scnt;
     split_aligned(count, nrings, align, soff, scnt);
     char* base = static_cast<char*>(t.data_ptr());
+
+    // hipGraph replay path: the composed multi-ring schedule is ~20 launches
+    // per ring; for stable buffers (gradient buckets, bench tensors) the
+    // whole schedule is captured once and replayed with ONE launch
+    // ("capture launch-bound inner loops in hipGraphs").
+    if (graphs_enabled_ && count * es >= graph_min_bytes_) {
+      GraphKey key{base, count, static_cast<int>(d), static_cast<int>(op),
+                   scale_alpha, nrings};
+      auto it = graphs_.find(key);
+      if (it != graphs_.end()) {
+        auto s0 = streams_[0];
+        TMA_HIP_CHECK(hipStreamWaitEvent(s0.stream(), ready, 0));
+        TMA_HIP_CHECK(hipGraphLaunch(it->second.exec, s0.stream()));
+        record_stream_uses(t, s0);
+        work->keep.push_back(it->second.scratch);
+        record_done(work, s0);
+        TMA_HIP_CHECK(hipEventDestroy(ready));
+        return work;
+      }
+    }
+
     for (int r = 0; r < nrings; ++r) {
       if (scnt[r] == 0) continue;
       auto s = streams_[r];
@@ -218,6 +251,12 @@ class RcclEngine {
       record_done(work, s);
     }
     TMA_HIP_CHECK(hipEventDestroy(ready));
+
+    // Capture for next time (after the eager run warmed all RCCL paths).
+    if (graphs_enabled_ && count * es >= graph_min_bytes_ &&
+        graphs_.size() < 64) {
+      try_capture_allreduce(base, count, d, op, scale_alpha, nrings, soff, scnt);
+    }
     return work;
   }
 
@@ -489,17 +528,15 @@ class RcclEngine {
   // indexed by GLOBAL RANK; the schedule walks them in ring order (at step s,
   // position p sends segment perm[(p-s-1) mod W]), so after W-1 steps rank g
   // owns the fully reduced segment g — on ANY ring permutation.
-  void ring_reduce_scatter(int r, char* base, size_t count, DType d, RedOp op,
-                           std::shared_ptr<GpuWork>& work) {
+  // `sc` is a double-buffer scratch of >= 2*ceil(count/W)*es bytes.
+  void ring_reduce_scatter_buf(int r, char* base, size_t count, DType d, RedOp op,
+                               char* sc) {
     const size_t es = dtype_size(d);
     const int W = world_;
     auto s = streams_[r];
     const auto& perm = rings_[r];
     std::vector<size_t> off, cnt;
     split_segments(count, W, off, cnt);
-    at::Tensor scratch = scratch_like(2 * cnt[0] * es, s);
-    work->keep.push_back(scratch);
-    char* sc = static_cast<char*>(scratch.data_ptr());
     const int p = pos_[r];
     for (int step = 0; step < W - 1; ++step) {
       int send_seg, recv_seg;
@@ -519,12 +556,33 @@ class RcclEngine {
     }
   }
 
+  // Eager wrappers: allocate per-call scratch from the caching allocator.
+  void ring_reduce_scatter(int r, char* base, size_t count, DType d, RedOp op,
+                           std::shared_ptr<GpuWork>& work) {
+    std::vector<size_t> off, cnt;
+    split_segments(count, world_, off, cnt);
+    at::Tensor scratch = scratch_like(2 * cnt[0] * dtype_size(d), streams_[r]);
+    work->keep.push_back(scratch);
+    ring_reduce_scatter_buf(r, base, count, d, op,
+                            static_cast<char*>(scratch.data_ptr()));
+  }
+
+  void ring_allreduce(int r, char* base, size_t count, DType d, RedOp op,
+                      std::shared_ptr<GpuWork>& work) {
+    std::vector<size_t> off, cnt;
+    split_segments(count, world_, off, cnt);
+    at::Tensor scratch = scratch_like(2 * cnt[0] * dtype_size(d), streams_[r]);
+    work->keep.push_back(scratch);
+    ring_allreduce_buf(r, base, count, d, op,
+                       static_cast<char*>(scratch.data_ptr()));
+  }
+
   // Full ring allreduce on ring r (reduce-scatter + allgather). After the RS
   // phase rank g owns segment g; the AG phase forwards owned segments around
   // the ring (at step s, position p sends segment perm[(p-s) mod W]).
-  void ring_allreduce(int r, char* base, size_t count, DType d, RedOp op,
-                      std::shared_ptr<GpuWork>& work) {
-    ring_reduce_scatter(r, base, count, d, op, work);
+  void ring_allreduce_buf(int r, char* base, size_t count, DType d, RedOp op,
+                          char* sc) {
+    ring_reduce_scatter_buf(r, base, count, d, op, sc);
     const size_t es = dtype_size(d);
     const int W = world_;
     auto s = streams_[r];
@@ -546,6 +604,100 @@ class RcclEngine {
     }
   }
 
+  struct GraphKey {
+    void* ptr;
+    size_t count;
+    int dtype, op;
+    double alpha;
+    int nrings;
+    bool operator==(const GraphKey& o) const {
+      return ptr == o.ptr && count == o.count && dtype == o.dtype &&
+             op == o.op && alpha == o.alpha && nrings == o.nrings;
+    }
+  };
+  struct GraphKeyHash {
+    size_t operator()(const GraphKey& k) const {
+      size_t h = std::hash<void*>()(k.ptr);
+      h = h * 31 + std::hash<size_t>()(k.count);
+      h = h * 31 + static_cast<size_t>(k.dtype * 7 + k.op * 131 + k.nrings);
+      h = h * 31 + std::hash<double>()(k.alpha);
+      return h;
+    }
+  };
+  struct GraphEntry {
+    hipGraphExec_t exec = nullptr;
+    at::Tensor scratch;
+    std::vector<hipEvent_t> events;
+  };
+
+  // Capture the full multi-ring schedule into a hipGraph; on any failure the
+  // graph path is disabled for the rest of the process (eager stays correct).
+  void try_capture_allreduce(char* base, size_t count, DType d, RedOp op,
+                             double alpha, int nrings,
+                             const std::vector<size_t>& soff,
+                             const std::vector<size_t>& scnt) {
+    const size_t es = dtype_size(d);
+    GraphEntry entry;
+    // Persistent per-entry scratch: nrings slices of 2*ceil(seg) each.
+    std::vector<size_t> off0, cnt0;
+    split_segments(scnt[0], world_, off0, cnt0);
+    size_t slice = 2 * cnt0[0] * es + 256;
+    {
+      at::Tensor scr = at::empty(
+          {static_cast<int64_t>(slice * nrings)},
+          at::TensorOptions().dtype(at::kByte).device(at::kCUDA, device_));
+      entry.scratch = scr;
+    }
+    auto s0 = streams_[0];
+    bool capturing = false;
+    try {
+      TMA_HIP_CHECK(hipStreamBeginCapture(s0.stream(),
+                                          hipStreamCaptureModeThreadLocal));
+      capturing = true;
+      hipEvent_t e0;
+      TMA_HIP_CHECK(hipEventCreateWithFlags(&e0, hipEventDisableTiming));
+      entry.events.push_back(e0);
+      TMA_HIP_CHECK(hipEventRecord(e0, s0.stream()));
+      for (int r = 1; r < nrings; ++r)
+        TMA_HIP_CHECK(hipStreamWaitEvent(streams_[r].stream(), e0, 0));
+      char* sc = static_cast<char*>(entry.scratch.data_ptr());
+      for (int r = 0; r < nrings; ++r) {
+        if (scnt[r] == 0) continue;
+        ring_allreduce_buf(r, base + soff[r] * es, scnt[r], d, op,
+                           sc + static_cast<size_t>(r) * slice);
+        if (alpha != 1.0)
+          launch_scale(base + soff[r] * es, scnt[r], d, alpha,
+                       streams_[r].stream());
+      }
+      for (int r = 1; r < nrings; ++r) {
+        hipEvent_t er;
+        TMA_HIP_CHECK(hipEventCreateWithFlags(&er, hipEventDisableTiming));
+        entry.events.push_back(er);
+        TMA_HIP_CHECK(hipEventRecord(er, streams_[r].stream()));
+        TMA_HIP_CHECK(hipStreamWaitEvent(s0.stream(), er, 0));
+      }
+      hipGraph_t graph = nullptr;
+      TMA_HIP_CHECK(hipStreamEndCapture(s0.stream(), &graph));
+      capturing = false;
+      TMA_HIP_CHECK(hipGraphInstantiate(&entry.exec, graph, nullptr, nullptr, 0));
+      TMA_HIP_CHECK(hipGraphDestroy(graph));
+      graphs_[GraphKey{base, count, static_cast<int>(d), static_cast<int>(op),
+                       alpha, nrings}] = entry;
+    } catch (const std::exception& e) {
+      if (capturing) {
+        hipGraph_t g = nullptr;
+        (void)hipStreamEndCapture(s0.stream(), &g);
+        if (g) (void)hipGraphDestroy(g);
+      }
+      for (auto ev : entry.events) (void)hipEventDestroy(ev);
+      graphs_enabled_ = false;  // never retry; eager path is always correct
+      if (std::getenv("TORCHMPI_LOG")) {
+        fprintf(stderr, "[torchmpi_amd] hipGraph capture disabled: %s\n",
+                e.what());
+      }
+    }
+  }
+
   int rank_, world_, device_;
   std::vector<std::vector<int>> rings_;
   std::vector<ncclComm_t> comms_;
@@ -553,6 +705,9 @@ class RcclEngine {
   std::vector<int> pos_, next_, prev_;
   size_t chunk_bytes_;
   size_t multiring_min_bytes_;
+  bool graphs_enabled_ = true;
+  size_t graph_min_bytes_ = 1u << 20;
+  std::unordered_map<GraphKey, GraphEntry, GraphKeyHash> graphs_;
 };
 
 // ---------------------------------------------------------------------------
@@ -675,6 +830,8 @@ void register_gpu(pybind11::module_& m) {
       .def("rank", &RcclEngine::rank)
       .def("size", &RcclEngine::size)
       .def("num_rings", &RcclEngine::num_rings)
+      .def("graphs_enabled", &RcclEngine::graphs_enabled)
+      .def("set_graphs_enabled", &RcclEngine::set_graphs_enabled)
       .def("allreduce", &RcclEngine::allreduce, py::arg("tensor"), py::arg("op"),
            py::arg("scale_alpha") = 1.0, py::call_guard<py::gil_scoped_release>())
       .def("reduce_scatter", &RcclEngine::reduce_scatter,
