@@ -123,3 +123,18 @@ def _self_sendrecv_body(tm):
 @requires_gpu
 def test_self_sendrecv():
     run_distributed(_self_sendrecv_body, world=1, use_gpu=True)
+
+
+def _graph_selfcheck_body(tm):
+    dev = torch.device("cuda", 0)
+    t = torch.full((1 << 18,), 1.0, device=dev)
+    ok = tm.communicator().engine.graph_selfcheck(t, 3)
+    assert ok, "hipGraph capture of RCCL ops failed on this stack"
+    # capture run does not execute; 3 replays double t three times: 1 -> 8
+    torch.cuda.synchronize()
+    assert torch.equal(t, torch.full((1 << 18,), 8.0, device=dev)), t[:4]
+
+
+@requires_gpu
+def test_hipgraph_rccl_capture_replay():
+    run_distributed(_graph_selfcheck_body, world=1, use_gpu=True)

@@ -194,6 +194,57 @@ class RcclEngine {
   bool graphs_enabled() const { return graphs_enabled_; }
   void set_graphs_enabled(bool v) { graphs_enabled_ = v; }
 
+  // 1-GPU validation of the hipGraph path (tests/test_gpu_collectives.py):
+  // captures {grouped self ncclSend/ncclRecv -> local reduce kernel} and
+  // replays it `replays` times: t doubles per replay. Exercises RCCL-ops-
+  // under-capture + instantiate + replay with the same machinery the
+  // multi-ring allreduce uses. Returns false (and disables graphs) if
+  // capture fails.
+  bool graph_selfcheck(at::Tensor t, int replays) {
+    check_gpu_tensor(t);
+    const size_t bytes = static_cast<size_t>(t.numel()) * t.element_size();
+    const DType d = dtype_of(t);
+    at::Tensor scratch = at::empty_like(t);
+    auto s0 = streams_[0];
+    hipEvent_t ready = record_ready();
+    TMA_HIP_CHECK(hipStreamWaitEvent(s0.stream(), ready, 0));
+    TMA_HIP_CHECK(hipEventDestroy(ready));
+    hipGraphExec_t exec = nullptr;
+    bool capturing = false;
+    try {
+      TMA_HIP_CHECK(hipStreamBeginCapture(s0.stream(),
+                                          hipStreamCaptureModeThreadLocal));
+      capturing = true;
+      TMA_NCCL_CHECK(ncclGroupStart());
+      TMA_NCCL_CHECK(ncclSend(t.data_ptr(), bytes, ncclInt8, rank_, comms_[0],
+                              s0.stream()));
+      TMA_NCCL_CHECK(ncclRecv(scratch.data_ptr(), bytes, ncclInt8, rank_,
+                              comms_[0], s0.stream()));
+      TMA_NCCL_CHECK(ncclGroupEnd());
+      launch_reduce(t.data_ptr(), scratch.data_ptr(), t.numel(), d, RedOp::SUM,
+                    s0.stream());
+      hipGraph_t g = nullptr;
+      TMA_HIP_CHECK(hipStreamEndCapture(s0.stream(), &g));
+      capturing = false;
+      TMA_HIP_CHECK(hipGraphInstantiate(&exec, g, nullptr, nullptr, 0));
+      TMA_HIP_CHECK(hipGraphDestroy(g));
+      for (int i = 0; i < replays; ++i)
+        TMA_HIP_CHECK(hipGraphLaunch(exec, s0.stream()));
+      TMA_HIP_CHECK(hipStreamSynchronize(s0.stream()));
+      (void)hipGraphExecDestroy(exec);
+      return true;
+    } catch (const std::exception&) {
+      if (capturing) {
+        hipGraph_t g = nullptr;
+        (void)hipStreamEndCapture(s0.stream(), &g);
+        if (g) (void)hipGraphDestroy(g);
+      }
+      if (exec) (void)hipGraphExecDestroy(exec);
+      graphs_enabled_ = false;
+      return false;
+    }
+  }
+
   // ---- verbs -------------------------------------------------------------
 
   std::shared_ptr<GpuWork> allreduce(at::Tensor t, RedOp op, double scale_alpha) {
@@ -832,6 +883,8 @@ void register_gpu(pybind11::module_& m) {
       .def("num_rings", &RcclEngine::num_rings)
       .def("graphs_enabled", &RcclEngine::graphs_enabled)
       .def("set_graphs_enabled", &RcclEngine::set_graphs_enabled)
+      .def("graph_selfcheck", &RcclEngine::graph_selfcheck,
+           py::call_guard<py::gil_scoped_release>())
       .def("allreduce", &RcclEngine::allreduce, py::arg("tensor"), py::arg("op"),
            py::arg("scale_alpha") = 1.0, py::call_guard<py::gil_scoped_release>())
       .def("reduce_scatter", &RcclEngine::reduce_scatter,
