@@ -88,3 +88,29 @@ def test_ring_count_matches_engine_env(monkeypatch):
     assert len(rings) == 2
     monkeypatch.delenv("TORCHMPI_NUM_RINGS")
     assert len(ring_permutations(8)) == 7
+
+
+def test_segment_parts_partition():
+    """Multi-ring verbs split every global segment into per-ring parts; the
+    parts must disjointly cover each segment and rank g's parts must be
+    contiguous (= its global segment)."""
+    for world in (2, 4, 8):
+        for total in (7, 64, 1000, 12345):
+            # global segments
+            base, rem = divmod(total, world)
+            off, cnt, o = [], [], 0
+            for g in range(world):
+                c = base + (1 if g < rem else 0)
+                off.append(o); cnt.append(c); o += c
+            for nrings in (1, 2, 7):
+                covered = [0] * world
+                prev_end = dict((g, off[g]) for g in range(world))
+                for r in range(nrings):
+                    off_r, cnt_r = _C.segment_parts(off, cnt, r, nrings, 4)
+                    for g in range(world):
+                        if cnt_r[g] == 0:
+                            continue
+                        assert off_r[g] == prev_end[g]  # contiguous parts
+                        prev_end[g] = off_r[g] + cnt_r[g]
+                        covered[g] += cnt_r[g]
+                assert covered == cnt, (world, total, nrings, covered, cnt)

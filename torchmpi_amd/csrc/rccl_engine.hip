@@ -332,17 +332,36 @@ class RcclEngine {
       return work;
     }
     hipEvent_t ready = record_ready();
-    auto s = streams_[0];
-    TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
     char* base = static_cast<char*>(input.data_ptr());
-    ring_reduce_scatter(0, base, count, d, op, work);
-    // After the RS phase rank g owns segment g (any ring permutation).
-    int own = rank_;
-    TMA_HIP_CHECK(hipMemcpyAsync(output.data_ptr(), base + off[own] * es,
-                                 cnt[own] * es, hipMemcpyDeviceToDevice, s.stream()));
-    record_stream_uses(input, s);
-    record_stream_uses(output, s);
-    record_done(work, s);
+    const size_t align = 16 / es;
+    int nrings = count * es >= multiring_min_bytes_ ? num_rings() : 1;
+    // Ring r reduces part r of every global segment; the union of rank g's
+    // parts is exactly its contiguous segment g.
+    for (int r = 0; r < nrings; ++r) {
+      std::vector<size_t> off_r, cnt_r;
+      segment_parts(off, cnt, r, nrings, align, off_r, cnt_r);
+      size_t maxc = 0;
+      for (auto c : cnt_r) maxc = std::max(maxc, c);
+      if (maxc == 0) continue;
+      auto s = streams_[r];
+      TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      at::Tensor scratch = scratch_like(2 * maxc * es, s);
+      work->keep.push_back(scratch);
+      ring_rs(r, base, off_r, cnt_r, d, op,
+              static_cast<char*>(scratch.data_ptr()));
+      record_stream_uses(input, s);
+      record_done(work, s);
+    }
+    // Copy the (now fully reduced) contiguous own segment into the output,
+    // after all rings finish: join rings on stream 0 via the work events.
+    auto s0 = streams_[0];
+    for (auto ev : work->events)
+      TMA_HIP_CHECK(hipStreamWaitEvent(s0.stream(), ev, 0));
+    TMA_HIP_CHECK(hipMemcpyAsync(output.data_ptr(), base + off[rank_] * es,
+                                 cnt[rank_] * es, hipMemcpyDeviceToDevice,
+                                 s0.stream()));
+    record_stream_uses(output, s0);
+    record_done(work, s0);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -357,16 +376,33 @@ class RcclEngine {
       return work;
     }
     hipEvent_t ready = record_ready();
-    auto s = streams_[0];
-    TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
-    // Chunk-pipelined chain in ring-0 order starting from root.
-    const auto& perm = rings_[0];
+    // Multi-chain: stripe the tensor over the link-disjoint rings; each
+    // stripe flows down ring r''s chain (root first) chunk-pipelined.
+    char* base = static_cast<char*>(t.data_ptr());
+    int nrings = bytes >= multiring_min_bytes_ ? num_rings() : 1;
+    std::vector<size_t> soff, scnt;
+    split_aligned(bytes, nrings, 16, soff, scnt);
+    for (int r = 0; r < nrings; ++r) {
+      if (scnt[r] == 0) continue;
+      auto s = streams_[r];
+      TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      chain_broadcast(r, base + soff[r], scnt[r], root);
+      record_stream_uses(t, s);
+      record_done(work, s);
+    }
+    TMA_HIP_CHECK(hipEventDestroy(ready));
+    return work;
+  }
+
+  // Chunk-pipelined chain broadcast of `bytes` at `base` over ring r.
+  void chain_broadcast(int r, char* base, size_t bytes, int root) {
+    auto s = streams_[r];
+    const auto& perm = rings_[r];
     int rootpos = 0;
     for (int i = 0; i < world_; ++i)
       if (perm[i] == root) rootpos = i;
-    int mypos = (pos_[0] - rootpos + world_) % world_;
-    int nxt = next_[0], prv = prev_[0];
-    char* base = static_cast<char*>(t.data_ptr());
+    int mypos = (pos_[r] - rootpos + world_) % world_;
+    int nxt = next_[r], prv = prev_[r];
     size_t nchunks = (bytes + chunk_bytes_ - 1) / chunk_bytes_;
     for (size_t c = 0; c <= nchunks; ++c) {
       bool do_recv = mypos > 0 && c < nchunks;
@@ -376,19 +412,15 @@ class RcclEngine {
       if (do_recv) {
         size_t o = c * chunk_bytes_;
         size_t n = std::min(chunk_bytes_, bytes - o);
-        TMA_NCCL_CHECK(ncclRecv(base + o, n, ncclInt8, prv, comms_[0], s.stream()));
+        TMA_NCCL_CHECK(ncclRecv(base + o, n, ncclInt8, prv, comms_[r], s.stream()));
       }
       if (do_send) {
         size_t o = (c - 1) * chunk_bytes_;
         size_t n = std::min(chunk_bytes_, bytes - o);
-        TMA_NCCL_CHECK(ncclSend(base + o, n, ncclInt8, nxt, comms_[0], s.stream()));
+        TMA_NCCL_CHECK(ncclSend(base + o, n, ncclInt8, nxt, comms_[r], s.stream()));
       }
       TMA_NCCL_CHECK(ncclGroupEnd());
     }
-    record_stream_uses(t, s);
-    record_done(work, s);
-    TMA_HIP_CHECK(hipEventDestroy(ready));
-    return work;
   }
 
   std::shared_ptr<GpuWork> reduce(at::Tensor t, RedOp op, int root) {
@@ -413,6 +445,7 @@ class RcclEngine {
     if (rank_ != root) {
       scratch = scratch_like(count * es, s);
       work->keep.push_back(scratch);
+      TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
       TMA_HIP_CHECK(hipMemcpyAsync(scratch.data_ptr(), base, count * es,
                                    hipMemcpyDeviceToDevice, s.stream()));
       base = static_cast<char*>(scratch.data_ptr());
@@ -460,24 +493,29 @@ class RcclEngine {
       return work;
     }
     hipEvent_t ready = record_ready();
-    auto s = streams_[0];
-    TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
-    int nxt = next_[0], prv = prev_[0];
-    const auto& perm = rings_[0];
-    // Ring forwarding in ring order: at step t we send the slot of the rank
-    // `t` positions behind us in the ring.
-    for (int t_ = 0; t_ < world_ - 1; ++t_) {
-      int send_owner = perm[(pos_[0] - t_ + world_) % world_];
-      int recv_owner = perm[(pos_[0] - t_ - 1 + 2 * world_) % world_];
-      TMA_NCCL_CHECK(ncclGroupStart());
-      TMA_NCCL_CHECK(ncclSend(ob + static_cast<size_t>(send_owner) * bytes, bytes,
-                              ncclInt8, nxt, comms_[0], s.stream()));
-      TMA_NCCL_CHECK(ncclRecv(ob + static_cast<size_t>(recv_owner) * bytes, bytes,
-                              ncclInt8, prv, comms_[0], s.stream()));
-      TMA_NCCL_CHECK(ncclGroupEnd());
+    // Multi-ring: slot g is a "global segment" owned by rank g; ring r
+    // forwards part r of every slot (ring_ag), using all xGMI links.
+    std::vector<size_t> off(world_), cnt(world_);
+    size_t elems = static_cast<size_t>(input.numel());
+    size_t es2 = static_cast<size_t>(input.element_size());
+    for (int g = 0; g < world_; ++g) {
+      off[g] = static_cast<size_t>(g) * elems;
+      cnt[g] = elems;
     }
-    record_stream_uses(output, s);
-    record_done(work, s);
+    const size_t align = 16 / es2 ? 16 / es2 : 1;
+    int nrings = bytes >= multiring_min_bytes_ ? num_rings() : 1;
+    for (int r = 0; r < nrings; ++r) {
+      std::vector<size_t> off_r, cnt_r;
+      segment_parts(off, cnt, r, nrings, align, off_r, cnt_r);
+      size_t maxc = 0;
+      for (auto c : cnt_r) maxc = std::max(maxc, c);
+      if (maxc == 0) continue;
+      auto s = streams_[r];
+      TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      ring_ag(r, ob, off_r, cnt_r, es2);
+      record_stream_uses(output, s);
+      record_done(work, s);
+    }
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -575,24 +613,42 @@ class RcclEngine {
     return t;
   }
 
-  // Ring reduce-scatter on ring r over [base, base + count*es). Segments are
-  // indexed by GLOBAL RANK; the schedule walks them in ring order (at step s,
-  // position p sends segment perm[(p-s-1) mod W]), so after W-1 steps rank g
-  // owns the fully reduced segment g — on ANY ring permutation.
-  // `sc` is a double-buffer scratch of >= 2*ceil(count/W)*es bytes.
-  void ring_reduce_scatter_buf(int r, char* base, size_t count, DType d, RedOp op,
-                               char* sc) {
+  // Per-ring segment tables for multi-ring verbs: the W global segments
+  // (indexed by rank) are each split into `nrings` aligned parts; ring r
+  // moves part r of every segment. Parts of one segment are contiguous, so
+  // the union over rings of rank g's parts is exactly global segment g.
+  void segment_parts(const std::vector<size_t>& off, const std::vector<size_t>& cnt,
+                     int r, int nrings, size_t align, std::vector<size_t>& off_r,
+                     std::vector<size_t>& cnt_r) {
+    const int W = static_cast<int>(off.size());
+    off_r.resize(W);
+    cnt_r.resize(W);
+    std::vector<size_t> po, pc;
+    for (int g = 0; g < W; ++g) {
+      split_aligned(cnt[g], nrings, align, po, pc);
+      off_r[g] = off[g] + po[r];
+      cnt_r[g] = pc[r];
+    }
+  }
+
+  // Ring reduce-scatter on ring r over the segment table (off/cnt indexed by
+  // GLOBAL RANK, elements). The schedule walks segments in ring order (at
+  // step s, position p sends segment perm[(p-s-1) mod W]), so after W-1
+  // steps rank g owns the fully reduced segment g — on ANY permutation.
+  // `sc` is a double-buffer scratch of >= 2*max(cnt)*es bytes.
+  void ring_rs(int r, char* base, const std::vector<size_t>& off,
+               const std::vector<size_t>& cnt, DType d, RedOp op, char* sc) {
     const size_t es = dtype_size(d);
     const int W = world_;
     auto s = streams_[r];
     const auto& perm = rings_[r];
-    std::vector<size_t> off, cnt;
-    split_segments(count, W, off, cnt);
+    size_t maxc = 0;
+    for (auto c : cnt) maxc = std::max(maxc, c);
     const int p = pos_[r];
     for (int step = 0; step < W - 1; ++step) {
       int send_seg, recv_seg;
       ring_step_segs(perm, W, p, step, /*phase=*/0, send_seg, recv_seg);
-      char* rbuf = sc + (step % 2) * cnt[0] * es;
+      char* rbuf = sc + (step % 2) * maxc * es;
       TMA_NCCL_CHECK(ncclGroupStart());
       if (cnt[send_seg] > 0)
         TMA_NCCL_CHECK(ncclSend(base + off[send_seg] * es, cnt[send_seg] * es,
@@ -605,6 +661,35 @@ class RcclEngine {
         launch_reduce(base + off[recv_seg] * es, rbuf, cnt[recv_seg], d, op,
                       s.stream());
     }
+  }
+
+  // Ring allgather on ring r over the segment table: rank g starts owning
+  // segment g; forwards owned segments around the ring.
+  void ring_ag(int r, char* base, const std::vector<size_t>& off,
+               const std::vector<size_t>& cnt, size_t es) {
+    const int W = world_;
+    auto s = streams_[r];
+    const auto& perm = rings_[r];
+    const int p = pos_[r];
+    for (int step = 0; step < W - 1; ++step) {
+      int send_seg, recv_seg;
+      ring_step_segs(perm, W, p, step, /*phase=*/1, send_seg, recv_seg);
+      TMA_NCCL_CHECK(ncclGroupStart());
+      if (cnt[send_seg] > 0)
+        TMA_NCCL_CHECK(ncclSend(base + off[send_seg] * es, cnt[send_seg] * es,
+                                ncclInt8, next_[r], comms_[r], s.stream()));
+      if (cnt[recv_seg] > 0)
+        TMA_NCCL_CHECK(ncclRecv(base + off[recv_seg] * es, cnt[recv_seg] * es,
+                                ncclInt8, prev_[r], comms_[r], s.stream()));
+      TMA_NCCL_CHECK(ncclGroupEnd());
+    }
+  }
+
+  void ring_reduce_scatter_buf(int r, char* base, size_t count, DType d, RedOp op,
+                               char* sc) {
+    std::vector<size_t> off, cnt;
+    split_segments(count, world_, off, cnt);
+    ring_rs(r, base, off, cnt, d, op, sc);
   }
 
   // Eager wrappers: allocate per-call scratch from the caching allocator.
@@ -628,31 +713,13 @@ class RcclEngine {
                        static_cast<char*>(scratch.data_ptr()));
   }
 
-  // Full ring allreduce on ring r (reduce-scatter + allgather). After the RS
-  // phase rank g owns segment g; the AG phase forwards owned segments around
-  // the ring (at step s, position p sends segment perm[(p-s) mod W]).
+  // Full ring allreduce on ring r (reduce-scatter + allgather).
   void ring_allreduce_buf(int r, char* base, size_t count, DType d, RedOp op,
                           char* sc) {
-    ring_reduce_scatter_buf(r, base, count, d, op, sc);
-    const size_t es = dtype_size(d);
-    const int W = world_;
-    auto s = streams_[r];
-    const auto& perm = rings_[r];
     std::vector<size_t> off, cnt;
-    split_segments(count, W, off, cnt);
-    const int p = pos_[r];
-    for (int step = 0; step < W - 1; ++step) {
-      int send_seg, recv_seg;
-      ring_step_segs(perm, W, p, step, /*phase=*/1, send_seg, recv_seg);
-      TMA_NCCL_CHECK(ncclGroupStart());
-      if (cnt[send_seg] > 0)
-        TMA_NCCL_CHECK(ncclSend(base + off[send_seg] * es, cnt[send_seg] * es,
-                                ncclInt8, next_[r], comms_[r], s.stream()));
-      if (cnt[recv_seg] > 0)
-        TMA_NCCL_CHECK(ncclRecv(base + off[recv_seg] * es, cnt[recv_seg] * es,
-                                ncclInt8, prev_[r], comms_[r], s.stream()));
-      TMA_NCCL_CHECK(ncclGroupEnd());
-    }
+    split_segments(count, world_, off, cnt);
+    ring_rs(r, base, off, cnt, d, op, sc);
+    ring_ag(r, base, off, cnt, dtype_size(d));
   }
 
   struct GraphKey {
@@ -902,6 +969,20 @@ void register_gpu(pybind11::module_& m) {
   m.def("nccl_unique_id", [] { return pybind11::bytes(py_get_unique_id()); });
   m.def("ring_schedule", &py_ring_schedule,
         "engine ring schedule for simulation tests");
+  m.def("segment_parts",
+        [](std::vector<size_t> off, std::vector<size_t> cnt, int r, int nrings,
+           size_t align) {
+          // mirrors RcclEngine::segment_parts (same split_aligned core)
+          std::vector<size_t> off_r(off.size()), cnt_r(off.size());
+          std::vector<size_t> po, pc;
+          for (size_t g = 0; g < off.size(); ++g) {
+            split_aligned(cnt[g], nrings, align, po, pc);
+            off_r[g] = off[g] + po[r];
+            cnt_r[g] = pc[r];
+          }
+          return std::make_pair(off_r, cnt_r);
+        },
+        "per-ring segment part tables (multi-ring verbs), for tests");
   m.def("reduce_into", &py_reduce_into, "dst = dst (op) src on current stream");
   m.def("scale_", &py_scale);
   m.def("build_pack_meta", &py_build_pack_meta);
