@@ -28,7 +28,7 @@ def make_model():
     )
 
 
-def main_worker():
+def main_worker(checkpoint=None, resume=False):
     import torchmpi_amd as tm
     from torchmpi_amd.nn import GradientSynchronizer, synchronize_parameters
 
@@ -38,9 +38,19 @@ def main_worker():
            if torch.cuda.is_available() else torch.device("cpu"))
 
     model = make_model().to(dev)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
+    start_epoch = 0
+    if resume and checkpoint and os.path.exists(checkpoint):
+        # Checkpoint/resume at the reference's capability level (SURVEY.md
+        # section 5): torch.save on rank 0 behind a barrier; every rank loads.
+        state = torch.load(checkpoint, map_location=dev, weights_only=False)
+        model.load_state_dict(state["model"])
+        opt.load_state_dict(state["opt"])
+        start_epoch = state["epoch"] + 1
+        if rank == 0:
+            print(f"resumed from {checkpoint} at epoch {start_epoch}")
     synchronize_parameters(model)          # replicas agree at t=0
     sync = GradientSynchronizer(model, average=True)
-    opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
 
     # Synthetic "dataset": 4096 samples, rank-sharded sample_index % W.
     torch.manual_seed(999)
@@ -48,7 +58,7 @@ def main_worker():
     labels = (data.view(4096, -1).sum(-1) * 10).long().abs() % 10
     my_idx = torch.arange(rank, 4096, W)
 
-    for epoch in range(3):
+    for epoch in range(start_epoch, 3):
         perm = torch.randperm(len(my_idx))
         total_loss, n = 0.0, 0
         for i in range(0, len(my_idx), 64):
@@ -63,24 +73,31 @@ def main_worker():
             n += len(batch)
         if rank == 0:
             print(f"epoch {epoch}: loss {total_loss / n:.4f}")
-    tm.barrier()
+        tm.barrier()               # grads/steps complete everywhere
+        if checkpoint and rank == 0:
+            torch.save({"model": model.state_dict(),
+                        "opt": opt.state_dict(), "epoch": epoch}, checkpoint)
+        tm.barrier()
     tm.stop()
 
 
-def _spawn(rank, world, port):
+def _spawn(rank, world, port, checkpoint=None, resume=False):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
                       MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
                       LOCAL_RANK=str(rank))
-    main_worker()
+    main_worker(checkpoint, resume)
 
 
 if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--world", type=int, default=0,
                     help="spawn N local processes (else use torchrun env)")
+    ap.add_argument("--checkpoint", type=str, default=None)
+    ap.add_argument("--resume", action="store_true")
     args = ap.parse_args()
     if args.world > 0:
         import torch.multiprocessing as mp
-        mp.spawn(_spawn, (args.world, 29610), nprocs=args.world, join=True)
+        mp.spawn(_spawn, (args.world, 29610, args.checkpoint, args.resume),
+                 nprocs=args.world, join=True)
     else:
-        main_worker()
+        main_worker(args.checkpoint, args.resume)

@@ -37,6 +37,7 @@ from .bootstrap import create_rendezvous, destroy_rendezvous
 from .parallel.communicator import Communicator
 from .parallel.hierarchical import hierarchical_allreduce
 from .utils import get_logger
+from .utils.trace import trace_dump, trace_enabled, trace_op
 
 RedOp = _C.RedOp
 
@@ -95,6 +96,7 @@ def stop() -> None:
     global _world, _comm_stack
     if _world is None:
         return
+    trace_dump()
     barrier()
     for c in _comm_stack:
         c.shutdown()
@@ -210,6 +212,22 @@ def _finish(h: Handle, async_op: bool):
     return h if async_op else h.wait()
 
 
+def _traced(fn):
+    """Chrome-trace wrapper (TORCHMPI_TRACE=prefix)."""
+    import functools
+
+    @functools.wraps(fn)
+    def wrapper(tensor, *a, **kw):
+        if not trace_enabled():
+            return fn(tensor, *a, **kw)
+        nbytes = (tensor.numel() * tensor.element_size()
+                  if hasattr(tensor, "numel") else 0)
+        with trace_op(fn.__name__, nbytes):
+            return fn(tensor, *a, **kw)
+    return wrapper
+
+
+@_traced
 def allreduce(tensor: torch.Tensor, op="sum", async_op: bool = False,
               average: bool = False, comm: Optional[Communicator] = None):
     """In-place allreduce. average=True fuses a 1/world scale (GPU: in the
@@ -229,6 +247,7 @@ def allreduce(tensor: torch.Tensor, op="sum", async_op: bool = False,
     return _finish(Handle(w, False, post), async_op)
 
 
+@_traced
 def reduce(tensor: torch.Tensor, op="sum", root: int = 0, async_op: bool = False,
            comm: Optional[Communicator] = None):
     """Reduce to `root` (in place at root; other ranks' tensors unchanged)."""
@@ -238,6 +257,7 @@ def reduce(tensor: torch.Tensor, op="sum", root: int = 0, async_op: bool = False
     return _finish(Handle(c.cpu.reduce(tensor, _op(op), root), False), async_op)
 
 
+@_traced
 def broadcast(tensor: torch.Tensor, root: int = 0, async_op: bool = False,
               comm: Optional[Communicator] = None):
     c = comm or communicator()
@@ -246,6 +266,7 @@ def broadcast(tensor: torch.Tensor, root: int = 0, async_op: bool = False,
     return _finish(Handle(c.cpu.broadcast(tensor, root), False), async_op)
 
 
+@_traced
 def allgather(tensor: torch.Tensor, out: Optional[torch.Tensor] = None,
               async_op: bool = False, comm: Optional[Communicator] = None):
     """Gather every rank's tensor; returns a tensor of shape
@@ -264,6 +285,7 @@ def allgather(tensor: torch.Tensor, out: Optional[torch.Tensor] = None,
     return out
 
 
+@_traced
 def reduce_scatter(tensor: torch.Tensor, out: torch.Tensor, op="sum",
                    async_op: bool = False, comm: Optional[Communicator] = None):
     """Reduce `tensor` across ranks and scatter segment `rank` into `out`.
@@ -276,6 +298,7 @@ def reduce_scatter(tensor: torch.Tensor, out: torch.Tensor, op="sum",
                    async_op)
 
 
+@_traced
 def sendreceive(send_tensor: torch.Tensor, dst: int, recv_tensor: torch.Tensor,
                 src: int, async_op: bool = False,
                 comm: Optional[Communicator] = None):
@@ -289,6 +312,7 @@ def sendreceive(send_tensor: torch.Tensor, dst: int, recv_tensor: torch.Tensor,
         Handle(c.cpu.sendrecv(send_tensor, dst, recv_tensor, src), False), async_op)
 
 
+@_traced
 def send(tensor: torch.Tensor, dst: int, async_op: bool = False,
          comm: Optional[Communicator] = None):
     c = comm or communicator()
@@ -297,6 +321,7 @@ def send(tensor: torch.Tensor, dst: int, async_op: bool = False,
     return _finish(Handle(c.cpu.send(tensor, dst), False), async_op)
 
 
+@_traced
 def recv(tensor: torch.Tensor, src: int, async_op: bool = False,
          comm: Optional[Communicator] = None):
     c = comm or communicator()
