@@ -157,3 +157,17 @@ def _barrier_body(tm):
 
 def test_barrier():
     run_distributed(_barrier_body, world=4)
+
+
+def _rhd_w8_body(tm):
+    # world=8 exercises all 3 levels of the recursive halving-doubling path
+    W, r = tm.size(), tm.rank()
+    for n in (8, 100, 1000, 4099):
+        t = torch.arange(n, dtype=torch.float32) + (r + 1)
+        tm.allreduce(t)
+        expect = torch.arange(n, dtype=torch.float32) * W + W * (W + 1) / 2
+        assert torch.allclose(t, expect), (n, t[:4], expect[:4])
+
+
+def test_allreduce_rhd_w8():
+    run_distributed(_rhd_w8_body, world=8)

@@ -30,6 +30,8 @@ CpuComm::CpuComm(int rank, const std::string& bind_addr) : rank_(rank), mesh_(bi
   const char* cb = std::getenv("TORCHMPI_CHUNK_BYTES");
   chunk_bytes_ = cb ? static_cast<size_t>(std::atoll(cb)) : (1u << 20);
   if (chunk_bytes_ < 4096) chunk_bytes_ = 4096;
+  const char* rm = std::getenv("TORCHMPI_RHD_MAX_BYTES");
+  rhd_max_bytes_ = rm ? static_cast<size_t>(std::atoll(rm)) : (256u << 10);
   worker_ = std::thread([this] { worker_loop(); });
 }
 
@@ -83,9 +85,70 @@ void CpuComm::worker_loop() {
 // Collectives
 // ---------------------------------------------------------------------------
 
+namespace {
+inline bool is_pow2(int x) { return x > 0 && (x & (x - 1)) == 0; }
+}  // namespace
+
+// Recursive halving-doubling allreduce: log2(W) exchange steps (vs the
+// ring's 2(W-1)) — the latency algorithm for small payloads. Requires W a
+// power of two. Segment ownership after the halving phase is bit-reversed,
+// mirrored back by the doubling phase.
+void CpuComm::allreduce_rhd(void* data, size_t count, DType d, RedOp op) {
+  const size_t es = dtype_size(d);
+  char* base = static_cast<char*>(data);
+  const int W = world_;
+  std::vector<size_t> off, cnt;
+  split_segments(count, W, off, cnt);
+  auto range_bytes = [&](int s0, int s1, size_t& o, size_t& n) {
+    o = off[s0] * es;
+    n = (off[s1 - 1] + cnt[s1 - 1] - off[s0]) * es;
+  };
+  std::vector<char> scratch(count * es);
+  int logw = 0;
+  while ((1 << logw) < W) ++logw;
+  int s0 = 0, s1 = W;
+  // Halving phase: keep one half, exchange + reduce.
+  for (int k = 0; k < logw; ++k) {
+    int partner = rank_ ^ (1 << k);
+    int mid = (s0 + s1) / 2;
+    int keep0, keep1, send0, send1;
+    if (((rank_ >> k) & 1) == 0) {
+      keep0 = s0; keep1 = mid; send0 = mid; send1 = s1;
+    } else {
+      keep0 = mid; keep1 = s1; send0 = s0; send1 = mid;
+    }
+    size_t so, sn, ko, kn;
+    range_bytes(send0, send1, so, sn);
+    range_bytes(keep0, keep1, ko, kn);
+    mesh_.duplex(partner, base + so, sn, partner, scratch.data(), kn);
+    if (kn > 0)
+      cpu_reduce(base + ko, scratch.data(), kn / es, d, op);
+    s0 = keep0; s1 = keep1;
+  }
+  // Doubling phase: mirror the exchanges back.
+  for (int k = logw - 1; k >= 0; --k) {
+    int partner = rank_ ^ (1 << k);
+    // partner's owned range at this level is the sibling of ours.
+    int width = s1 - s0;
+    int sib0 = (s0 % (2 * width) == 0) ? s0 + width : s0 - width;
+    int sib1 = sib0 + width;
+    size_t mo, mn, po, pn;
+    range_bytes(s0, s1, mo, mn);
+    range_bytes(sib0, sib1, po, pn);
+    mesh_.duplex(partner, base + mo, mn, partner, base + po, pn);
+    s0 = std::min(s0, sib0);
+    s1 = std::max(s1, sib1);
+  }
+}
+
 void CpuComm::allreduce(void* data, size_t count, DType d, RedOp op) {
   if (world_ == 1 || count == 0) return;
   const size_t es = dtype_size(d);
+  if (is_pow2(world_) && count * es <= rhd_max_bytes_ &&
+      count >= static_cast<size_t>(world_)) {
+    allreduce_rhd(data, count, d, op);
+    return;
+  }
   char* base = static_cast<char*>(data);
   const int W = world_;
   const int right = (rank_ + 1) % W, left = (rank_ - 1 + W) % W;

@@ -50,6 +50,7 @@ class CpuComm {
 
   // --- blocking verbs (call only from the worker thread via enqueue) ---
   void allreduce(void* data, size_t count, DType d, RedOp op);
+  void allreduce_rhd(void* data, size_t count, DType d, RedOp op);
   void reduce_scatter(void* data, void* out, size_t total_count, DType d, RedOp op);
   void broadcast(void* data, size_t bytes, int root);
   void reduce(void* data, size_t count, DType d, RedOp op, int root);
@@ -72,6 +73,7 @@ class CpuComm {
   int world_ = 1;
   TcpMesh mesh_;
   size_t chunk_bytes_;
+  size_t rhd_max_bytes_;
 
   struct Task {
     std::function<void()> fn;

@@ -178,6 +178,8 @@ class RcclEngine {
     graphs_enabled_ = !(ge && std::string(ge) == "0");
     const char* gm = std::getenv("TORCHMPI_GRAPH_MIN_BYTES");
     if (gm) graph_min_bytes_ = static_cast<size_t>(std::atoll(gm));
+    const char* rh = std::getenv("TORCHMPI_RHD_MAX_BYTES");
+    rhd_max_bytes_ = rh ? static_cast<size_t>(std::atoll(rh)) : (256u << 10);
   }
 
   ~RcclEngine() {
@@ -264,6 +266,24 @@ class RcclEngine {
     }
 
     hipEvent_t ready = record_ready();
+
+    // Small payloads: recursive halving-doubling (log2 W latency).
+    if ((world_ & (world_ - 1)) == 0 && count * es <= rhd_max_bytes_ &&
+        count >= static_cast<size_t>(world_)) {
+      auto s = streams_[0];
+      TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      at::Tensor scratch = scratch_like(count * es, s);
+      work->keep.push_back(scratch);
+      rhd_allreduce(static_cast<char*>(t.data_ptr()), count, d, op,
+                    static_cast<char*>(scratch.data_ptr()));
+      if (scale_alpha != 1.0)
+        launch_scale(t.data_ptr(), count, d, scale_alpha, s.stream());
+      record_stream_uses(t, s);
+      record_done(work, s);
+      TMA_HIP_CHECK(hipEventDestroy(ready));
+      return work;
+    }
+
     const size_t align = 16 / es;  // keep 16B alignment for the reduce kernel
     int nrings = count * es >= multiring_min_bytes_ ? num_rings() : 1;
 
@@ -692,6 +712,70 @@ class RcclEngine {
     ring_rs(r, base, off, cnt, d, op, sc);
   }
 
+  // Recursive halving-doubling allreduce on comms_[0]/streams_[0]:
+  // log2(W) exchange steps (vs the ring's 2(W-1)) — the latency algorithm
+  // for small payloads (north star: "ring and recursive-halving allreduce").
+  // Same schedule as CpuComm::allreduce_rhd (tests cover it at W=2/4/8 over
+  // the TCP transport). Requires W a power of two; `sc` >= count*es bytes.
+  void rhd_allreduce(char* base, size_t count, DType d, RedOp op, char* sc) {
+    const size_t es = dtype_size(d);
+    const int W = world_;
+    auto s = streams_[0];
+    std::vector<size_t> off, cnt;
+    split_segments(count, W, off, cnt);
+    auto range_bytes = [&](int a, int b, size_t& o, size_t& n) {
+      o = off[a] * es;
+      n = (off[b - 1] + cnt[b - 1] - off[a]) * es;
+    };
+    int logw = 0;
+    while ((1 << logw) < W) ++logw;
+    int s0 = 0, s1 = W;
+    for (int k = 0; k < logw; ++k) {
+      int partner = rank_ ^ (1 << k);
+      int mid = (s0 + s1) / 2;
+      int keep0, keep1, send0, send1;
+      if (((rank_ >> k) & 1) == 0) {
+        keep0 = s0; keep1 = mid; send0 = mid; send1 = s1;
+      } else {
+        keep0 = mid; keep1 = s1; send0 = s0; send1 = mid;
+      }
+      size_t so, sn, ko, kn;
+      range_bytes(send0, send1, so, sn);
+      range_bytes(keep0, keep1, ko, kn);
+      TMA_NCCL_CHECK(ncclGroupStart());
+      if (sn > 0)
+        TMA_NCCL_CHECK(ncclSend(base + so, sn, ncclInt8, partner, comms_[0],
+                                s.stream()));
+      if (kn > 0)
+        TMA_NCCL_CHECK(ncclRecv(sc, kn, ncclInt8, partner, comms_[0],
+                                s.stream()));
+      TMA_NCCL_CHECK(ncclGroupEnd());
+      if (kn > 0)
+        launch_reduce(base + ko, sc, kn / es, d, op, s.stream());
+      s0 = keep0;
+      s1 = keep1;
+    }
+    for (int k = logw - 1; k >= 0; --k) {
+      int partner = rank_ ^ (1 << k);
+      int width = s1 - s0;
+      int sib0 = (s0 % (2 * width) == 0) ? s0 + width : s0 - width;
+      int sib1 = sib0 + width;
+      size_t mo, mn, po, pn;
+      range_bytes(s0, s1, mo, mn);
+      range_bytes(sib0, sib1, po, pn);
+      TMA_NCCL_CHECK(ncclGroupStart());
+      if (mn > 0)
+        TMA_NCCL_CHECK(ncclSend(base + mo, mn, ncclInt8, partner, comms_[0],
+                                s.stream()));
+      if (pn > 0)
+        TMA_NCCL_CHECK(ncclRecv(base + po, pn, ncclInt8, partner, comms_[0],
+                                s.stream()));
+      TMA_NCCL_CHECK(ncclGroupEnd());
+      s0 = std::min(s0, sib0);
+      s1 = std::max(s1, sib1);
+    }
+  }
+
   // Eager wrappers: allocate per-call scratch from the caching allocator.
   void ring_reduce_scatter(int r, char* base, size_t count, DType d, RedOp op,
                            std::shared_ptr<GpuWork>& work) {
@@ -823,6 +907,7 @@ class RcclEngine {
   std::vector<int> pos_, next_, prev_;
   size_t chunk_bytes_;
   size_t multiring_min_bytes_;
+  size_t rhd_max_bytes_ = 256u << 10;
   bool graphs_enabled_ = true;
   size_t graph_min_bytes_ = 1u << 20;
   std::unordered_map<GraphKey, GraphEntry, GraphKeyHash> graphs_;
