@@ -138,3 +138,23 @@ def _graph_selfcheck_body(tm):
 @requires_gpu
 def test_hipgraph_rccl_capture_replay():
     run_distributed(_graph_selfcheck_body, world=1, use_gpu=True)
+
+
+def _fan_verbs_world1_body(tm):
+    dev = torch.device("cuda", 0)
+    t = torch.arange(100, device=dev, dtype=torch.float32)
+    out = tm.gather(t, root=0)
+    torch.cuda.synchronize()
+    assert out.shape == (1, 100) and torch.equal(out[0], t)
+    dst = torch.zeros(100, device=dev)
+    tm.scatter(out.view(-1), dst, root=0)
+    torch.cuda.synchronize()
+    assert torch.equal(dst, t)
+    a2a = tm.alltoall(t)
+    torch.cuda.synchronize()
+    assert torch.equal(a2a, t)
+
+
+@requires_gpu
+def test_fan_verbs_world1():
+    run_distributed(_fan_verbs_world1_body, world=1, use_gpu=True)

@@ -333,24 +333,28 @@ def recv(tensor: torch.Tensor, src: int, async_op: bool = False,
 def gather(tensor: torch.Tensor, root: int = 0, out: Optional[torch.Tensor] = None,
            comm: Optional[Communicator] = None):
     c = comm or communicator()
-    if tensor.is_cuda:
-        raise NotImplementedError("gather: use allgather on GPU")
     if c.rank() == root and out is None:
         out = torch.empty((c.size(),) + tuple(tensor.shape), dtype=tensor.dtype,
                           device=tensor.device)
-    if out is None:
-        out = torch.empty(0, dtype=tensor.dtype)
-    Handle(c.cpu.gather(tensor, out, root), False).wait()
+    if tensor.is_cuda:
+        if out is None:
+            out = torch.empty(0, dtype=tensor.dtype, device=tensor.device)
+        Handle(c.engine.gather(tensor, out, root), True).wait()
+    else:
+        if out is None:
+            out = torch.empty(0, dtype=tensor.dtype)
+        Handle(c.cpu.gather(tensor, out, root), False).wait()
     return out if c.rank() == root else None
 
 
 def scatter(tensor_in: Optional[torch.Tensor], out: torch.Tensor, root: int = 0,
             comm: Optional[Communicator] = None):
     c = comm or communicator()
-    if out.is_cuda:
-        raise NotImplementedError("scatter: GPU path not implemented yet")
     if tensor_in is None:
-        tensor_in = torch.empty(0, dtype=out.dtype)
+        tensor_in = torch.empty(0, dtype=out.dtype, device=out.device)
+    if out.is_cuda:
+        Handle(c.engine.scatter(tensor_in, out, root), True).wait()
+        return out
     Handle(c.cpu.scatter(tensor_in, out, root), False).wait()
     return out
 
@@ -358,10 +362,11 @@ def scatter(tensor_in: Optional[torch.Tensor], out: torch.Tensor, root: int = 0,
 def alltoall(tensor: torch.Tensor, out: Optional[torch.Tensor] = None,
              comm: Optional[Communicator] = None):
     c = comm or communicator()
-    if tensor.is_cuda:
-        raise NotImplementedError("alltoall: GPU path not implemented yet")
     if out is None:
         out = torch.empty_like(tensor)
+    if tensor.is_cuda:
+        Handle(c.engine.alltoall(tensor, out), True).wait()
+        return out
     Handle(c.cpu.alltoall(tensor, out), False).wait()
     return out
 

@@ -564,6 +564,102 @@ class RcclEngine {
     return work;
   }
 
+  // Gather to root / scatter from root / all-to-all: grouped p2p fans
+  // (reference MPI_Gather/Scatter/Alltoall verb parity; xGMI is fully
+  // connected so every pair is one direct link).
+  std::shared_ptr<GpuWork> gather(at::Tensor input, at::Tensor output, int root) {
+    check_gpu_tensor(input);
+    auto work = std::make_shared<GpuWork>();
+    work->keep = {input, output};
+    size_t bytes = static_cast<size_t>(input.numel()) * input.element_size();
+    hipEvent_t ready = record_ready();
+    auto s = streams_[0];
+    TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    TMA_NCCL_CHECK(ncclGroupStart());
+    if (rank_ == root) {
+      check_gpu_tensor(output);
+      char* ob = static_cast<char*>(output.data_ptr());
+      TMA_HIP_CHECK(hipMemcpyAsync(ob + static_cast<size_t>(root) * bytes,
+                                   input.data_ptr(), bytes,
+                                   hipMemcpyDeviceToDevice, s.stream()));
+      for (int r = 0; r < world_; ++r)
+        if (r != root)
+          TMA_NCCL_CHECK(ncclRecv(ob + static_cast<size_t>(r) * bytes, bytes,
+                                  ncclInt8, r, comms_[0], s.stream()));
+      record_stream_uses(output, s);
+    } else {
+      TMA_NCCL_CHECK(
+          ncclSend(input.data_ptr(), bytes, ncclInt8, root, comms_[0], s.stream()));
+    }
+    TMA_NCCL_CHECK(ncclGroupEnd());
+    record_stream_uses(input, s);
+    record_done(work, s);
+    TMA_HIP_CHECK(hipEventDestroy(ready));
+    return work;
+  }
+
+  std::shared_ptr<GpuWork> scatter(at::Tensor input, at::Tensor output, int root) {
+    check_gpu_tensor(output);
+    auto work = std::make_shared<GpuWork>();
+    work->keep = {input, output};
+    size_t bytes = static_cast<size_t>(output.numel()) * output.element_size();
+    hipEvent_t ready = record_ready();
+    auto s = streams_[0];
+    TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    TMA_NCCL_CHECK(ncclGroupStart());
+    if (rank_ == root) {
+      check_gpu_tensor(input);
+      char* ib = static_cast<char*>(input.data_ptr());
+      TMA_HIP_CHECK(hipMemcpyAsync(output.data_ptr(),
+                                   ib + static_cast<size_t>(root) * bytes, bytes,
+                                   hipMemcpyDeviceToDevice, s.stream()));
+      for (int r = 0; r < world_; ++r)
+        if (r != root)
+          TMA_NCCL_CHECK(ncclSend(ib + static_cast<size_t>(r) * bytes, bytes,
+                                  ncclInt8, r, comms_[0], s.stream()));
+      record_stream_uses(input, s);
+    } else {
+      TMA_NCCL_CHECK(ncclRecv(output.data_ptr(), bytes, ncclInt8, root, comms_[0],
+                              s.stream()));
+    }
+    TMA_NCCL_CHECK(ncclGroupEnd());
+    record_stream_uses(output, s);
+    record_done(work, s);
+    TMA_HIP_CHECK(hipEventDestroy(ready));
+    return work;
+  }
+
+  std::shared_ptr<GpuWork> alltoall(at::Tensor input, at::Tensor output) {
+    check_gpu_tensor(input);
+    check_gpu_tensor(output);
+    auto work = std::make_shared<GpuWork>();
+    work->keep = {input, output};
+    size_t bytes =
+        static_cast<size_t>(input.numel()) * input.element_size() / world_;
+    hipEvent_t ready = record_ready();
+    auto s = streams_[0];
+    TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    const char* ib = static_cast<const char*>(input.data_ptr());
+    char* ob = static_cast<char*>(output.data_ptr());
+    TMA_HIP_CHECK(hipMemcpyAsync(ob + static_cast<size_t>(rank_) * bytes,
+                                 ib + static_cast<size_t>(rank_) * bytes, bytes,
+                                 hipMemcpyDeviceToDevice, s.stream()));
+    TMA_NCCL_CHECK(ncclGroupStart());
+    for (int r = 0; r < world_; ++r) {
+      if (r == rank_) continue;
+      TMA_NCCL_CHECK(ncclSend(ib + static_cast<size_t>(r) * bytes, bytes, ncclInt8,
+                              r, comms_[0], s.stream()));
+      TMA_NCCL_CHECK(ncclRecv(ob + static_cast<size_t>(r) * bytes, bytes, ncclInt8,
+                              r, comms_[0], s.stream()));
+    }
+    TMA_NCCL_CHECK(ncclGroupEnd());
+    record_stream_uses(input, s);
+    record_stream_uses(output, s);
+    record_done(work, s);
+    TMA_HIP_CHECK(hipEventDestroy(ready));
+    return work;
+  }
+
   std::shared_ptr<GpuWork> send(at::Tensor t, int dst) {
     check_gpu_tensor(t);
     auto work = std::make_shared<GpuWork>();
@@ -1047,6 +1143,12 @@ void register_gpu(pybind11::module_& m) {
       .def("allgather", &RcclEngine::allgather,
            py::call_guard<py::gil_scoped_release>())
       .def("sendrecv", &RcclEngine::sendrecv,
+           py::call_guard<py::gil_scoped_release>())
+      .def("gather", &RcclEngine::gather,
+           py::call_guard<py::gil_scoped_release>())
+      .def("scatter", &RcclEngine::scatter,
+           py::call_guard<py::gil_scoped_release>())
+      .def("alltoall", &RcclEngine::alltoall,
            py::call_guard<py::gil_scoped_release>())
       .def("send", &RcclEngine::send, py::call_guard<py::gil_scoped_release>())
       .def("recv", &RcclEngine::recv, py::call_guard<py::gil_scoped_release>());
