@@ -263,7 +263,11 @@ void PsContext::loop() {
           handle_frame(peer_of[i]);
         } catch (const std::exception& e) {
           if (!running_) return;
-          throw;
+          // Never std::terminate the process from the event loop: log and
+          // stop serving (pending client Works will time out loudly).
+          fprintf(stderr, "[torchmpi_amd ps] event loop error: %s\n", e.what());
+          running_ = false;
+          return;
         }
       }
     }

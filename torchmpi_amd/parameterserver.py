@@ -94,6 +94,10 @@ class ParameterServer:
             self._ctx.register_tensor(tid, flat)
             self._kind[tid] = "cpu"
         self._tids[id(tensor)] = tid
+        # Registration is collective: no client may push before every owner
+        # has its shard store (reference: PS init posts persistent receives
+        # before any traffic).
+        self.comm.barrier()
         return tid
 
     def _tid_of(self, tensor: torch.Tensor) -> int:
@@ -139,6 +143,9 @@ class ParameterServer:
     prefetch = receive  # reference: prefetch == async receive
 
     def free(self, tensor: torch.Tensor) -> None:
+        """Collective: all ranks free together (a live client fetching a
+        freed shard would kill the owner's event loop)."""
+        self.comm.barrier()
         tid = self._tids.pop(id(tensor), None)
         if tid is not None:
             if self._kind.get(tid) == "gpu":
@@ -148,5 +155,9 @@ class ParameterServer:
             self._kind.pop(tid, None)
 
     def shutdown(self):
+        """Collective: barrier FIRST (with the event loop still serving), so
+        a fast rank cannot stop serving while a slow rank's request is in
+        flight (each client waits its own handles before calling this)."""
+        self.comm.barrier()
         self._ctx.stop()
         self._gpu = None
