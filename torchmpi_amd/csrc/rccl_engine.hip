@@ -29,6 +29,7 @@
 #include <rccl/rccl.h>
 
 #include <cstdio>
+#include <map>
 #include <memory>
 #include <unordered_map>
 #include <vector>
@@ -183,6 +184,8 @@ class RcclEngine {
   }
 
   ~RcclEngine() {
+    if (order_ev_) (void)hipEventDestroy(order_ev_);
+    if (join_ev_) (void)hipEventDestroy(join_ev_);
     for (auto& kv : graphs_) {
       if (kv.second.exec) (void)hipGraphExecDestroy(kv.second.exec);
       for (auto ev : kv.second.events) (void)hipEventDestroy(ev);
@@ -272,6 +275,7 @@ class RcclEngine {
         count >= static_cast<size_t>(world_)) {
       auto s = streams_[0];
       TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      order_in(s);
       at::Tensor scratch = scratch_like(count * es, s);
       work->keep.push_back(scratch);
       rhd_allreduce(static_cast<char*>(t.data_ptr()), count, d, op,
@@ -280,6 +284,7 @@ class RcclEngine {
         launch_scale(t.data_ptr(), count, d, scale_alpha, s.stream());
       record_stream_uses(t, s);
       record_done(work, s);
+      order_out(1);
       TMA_HIP_CHECK(hipEventDestroy(ready));
       return work;
     }
@@ -302,10 +307,12 @@ class RcclEngine {
       if (it != graphs_.end()) {
         auto s0 = streams_[0];
         TMA_HIP_CHECK(hipStreamWaitEvent(s0.stream(), ready, 0));
+        order_in(s0);
         TMA_HIP_CHECK(hipGraphLaunch(it->second.exec, s0.stream()));
         record_stream_uses(t, s0);
         work->keep.push_back(it->second.scratch);
         record_done(work, s0);
+        order_out(1);
         TMA_HIP_CHECK(hipEventDestroy(ready));
         return work;
       }
@@ -315,17 +322,19 @@ class RcclEngine {
       if (scnt[r] == 0) continue;
       auto s = streams_[r];
       TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      order_in(s);
       ring_allreduce(r, base + soff[r] * es, scnt[r], d, op, work);
       if (scale_alpha != 1.0)
         launch_scale(base + soff[r] * es, scnt[r], d, scale_alpha, s.stream());
       record_stream_uses(t, s);
       record_done(work, s);
     }
+    order_out(nrings);
     TMA_HIP_CHECK(hipEventDestroy(ready));
 
     // Capture for next time (after the eager run warmed all RCCL paths).
     if (graphs_enabled_ && count * es >= graph_min_bytes_ &&
-        graphs_.size() < 64) {
+        graphs_.size() < 1024) {
       try_capture_allreduce(base, count, d, op, scale_alpha, nrings, soff, scnt);
     }
     return work;
@@ -365,6 +374,7 @@ class RcclEngine {
       if (maxc == 0) continue;
       auto s = streams_[r];
       TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      order_in(s);
       at::Tensor scratch = scratch_like(2 * maxc * es, s);
       work->keep.push_back(scratch);
       ring_rs(r, base, off_r, cnt_r, d, op,
@@ -382,6 +392,7 @@ class RcclEngine {
                                  s0.stream()));
     record_stream_uses(output, s0);
     record_done(work, s0);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -406,10 +417,12 @@ class RcclEngine {
       if (scnt[r] == 0) continue;
       auto s = streams_[r];
       TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      order_in(s);
       chain_broadcast(r, base + soff[r], scnt[r], root);
       record_stream_uses(t, s);
       record_done(work, s);
     }
+    order_out(nrings);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -457,6 +470,7 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     auto s = streams_[0];
     TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    order_in(s);
 
     // Non-root ranks work on a scratch copy so their input is preserved
     // (MPI_Reduce semantics, reference lib/collectives.cpp reduce).
@@ -490,6 +504,7 @@ class RcclEngine {
     TMA_NCCL_CHECK(ncclGroupEnd());
     record_stream_uses(t, s);
     record_done(work, s);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -532,10 +547,12 @@ class RcclEngine {
       if (maxc == 0) continue;
       auto s = streams_[r];
       TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+      order_in(s);
       ring_ag(r, ob, off_r, cnt_r, es2);
       record_stream_uses(output, s);
       record_done(work, s);
     }
+    order_out(nrings);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -549,6 +566,7 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     auto s = streams_[0];
     TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    order_in(s);
     size_t sbytes = static_cast<size_t>(send_t.numel()) * send_t.element_size();
     size_t rbytes = static_cast<size_t>(recv_t.numel()) * recv_t.element_size();
     TMA_NCCL_CHECK(ncclGroupStart());
@@ -560,6 +578,7 @@ class RcclEngine {
     record_stream_uses(send_t, s);
     record_stream_uses(recv_t, s);
     record_done(work, s);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -575,6 +594,7 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     auto s = streams_[0];
     TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    order_in(s);
     TMA_NCCL_CHECK(ncclGroupStart());
     if (rank_ == root) {
       check_gpu_tensor(output);
@@ -594,6 +614,7 @@ class RcclEngine {
     TMA_NCCL_CHECK(ncclGroupEnd());
     record_stream_uses(input, s);
     record_done(work, s);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -606,6 +627,7 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     auto s = streams_[0];
     TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    order_in(s);
     TMA_NCCL_CHECK(ncclGroupStart());
     if (rank_ == root) {
       check_gpu_tensor(input);
@@ -625,6 +647,7 @@ class RcclEngine {
     TMA_NCCL_CHECK(ncclGroupEnd());
     record_stream_uses(output, s);
     record_done(work, s);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -639,6 +662,7 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     auto s = streams_[0];
     TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    order_in(s);
     const char* ib = static_cast<const char*>(input.data_ptr());
     char* ob = static_cast<char*>(output.data_ptr());
     TMA_HIP_CHECK(hipMemcpyAsync(ob + static_cast<size_t>(rank_) * bytes,
@@ -656,6 +680,7 @@ class RcclEngine {
     record_stream_uses(input, s);
     record_stream_uses(output, s);
     record_done(work, s);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -667,11 +692,13 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     auto s = streams_[0];
     TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    order_in(s);
     TMA_NCCL_CHECK(ncclSend(t.data_ptr(),
                             static_cast<size_t>(t.numel()) * t.element_size(), ncclInt8,
                             dst, comms_[0], s.stream()));
     record_stream_uses(t, s);
     record_done(work, s);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -683,11 +710,13 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     auto s = streams_[0];
     TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
+    order_in(s);
     TMA_NCCL_CHECK(ncclRecv(t.data_ptr(),
                             static_cast<size_t>(t.numel()) * t.element_size(), ncclInt8,
                             src, comms_[0], s.stream()));
     record_stream_uses(t, s);
     record_done(work, s);
+    order_out(1);
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -715,6 +744,31 @@ class RcclEngine {
     TMA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
     TMA_HIP_CHECK(hipEventRecord(ev, c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream()));
     w->events.push_back(ev);
+  }
+
+  // ---- inter-collective ordering -----------------------------------------
+  // Collectives on this engine must execute in issue order on every rank
+  // (grouped-p2p matching is per-communicator call order). Per-stream FIFO
+  // gives that while everything is eager, but hipGraph replays root/join
+  // only on streams_[0]; order_in/order_out chain every verb through one
+  // persistent event so replays and eager multi-stream ops never overlap on
+  // a communicator.
+  void order_in(c10::hip::HIPStreamMasqueradingAsCUDA s) {
+    if (order_ev_) TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), order_ev_, 0));
+  }
+
+  // Join streams [1, nrings) into streams_[0] and mark the op boundary.
+  void order_out(int nrings_used) {
+    if (!order_ev_)
+      TMA_HIP_CHECK(hipEventCreateWithFlags(&order_ev_, hipEventDisableTiming));
+    auto s0 = streams_[0];
+    for (int r = 1; r < nrings_used; ++r) {
+      if (!join_ev_)
+        TMA_HIP_CHECK(hipEventCreateWithFlags(&join_ev_, hipEventDisableTiming));
+      TMA_HIP_CHECK(hipEventRecord(join_ev_, streams_[r].stream()));
+      TMA_HIP_CHECK(hipStreamWaitEvent(s0.stream(), join_ev_, 0));
+    }
+    TMA_HIP_CHECK(hipEventRecord(order_ev_, s0.stream()));
   }
 
   void record_stream_uses(const at::Tensor& t, c10::hip::HIPStreamMasqueradingAsCUDA s) {
@@ -936,16 +990,23 @@ class RcclEngine {
                              const std::vector<size_t>& scnt) {
     const size_t es = dtype_size(d);
     GraphEntry entry;
-    // Persistent per-entry scratch: nrings slices of 2*ceil(seg) each.
+    // Graph-captured pointers must stay valid forever, but replays of
+    // different graphs serialize on the same comm streams, so entries can
+    // SHARE scratch: one persistent buffer per power-of-2 size class.
     std::vector<size_t> off0, cnt0;
     split_segments(scnt[0], world_, off0, cnt0);
     size_t slice = 2 * cnt0[0] * es + 256;
-    {
+    size_t need = slice * nrings;
+    size_t cls = 4096;
+    while (cls < need) cls <<= 1;
+    auto pit = graph_scratch_pool_.find(cls);
+    if (pit == graph_scratch_pool_.end()) {
       at::Tensor scr = at::empty(
-          {static_cast<int64_t>(slice * nrings)},
+          {static_cast<int64_t>(cls)},
           at::TensorOptions().dtype(at::kByte).device(at::kCUDA, device_));
-      entry.scratch = scr;
+      pit = graph_scratch_pool_.emplace(cls, scr).first;
     }
+    entry.scratch = pit->second;
     auto s0 = streams_[0];
     bool capturing = false;
     try {
@@ -1001,12 +1062,15 @@ class RcclEngine {
   std::vector<ncclComm_t> comms_;
   std::vector<c10::hip::HIPStreamMasqueradingAsCUDA> streams_;
   std::vector<int> pos_, next_, prev_;
+  hipEvent_t order_ev_ = nullptr;
+  hipEvent_t join_ev_ = nullptr;
   size_t chunk_bytes_;
   size_t multiring_min_bytes_;
   size_t rhd_max_bytes_ = 256u << 10;
   bool graphs_enabled_ = true;
   size_t graph_min_bytes_ = 1u << 20;
   std::unordered_map<GraphKey, GraphEntry, GraphKeyHash> graphs_;
+  std::map<size_t, at::Tensor> graph_scratch_pool_;
 };
 
 // ---------------------------------------------------------------------------
