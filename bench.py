@@ -108,6 +108,9 @@ def main():
     # (NCHW bf16 falls back to naive_conv — 81% of step time, see profiles/).
     model = resnet50().to(device=device)
     if use_gpu:
+        if os.environ.get("TORCHMPI_FUSED_BN", "1") != "0":
+            from torchmpi_amd.ops.batchnorm import convert_batchnorm
+            convert_batchnorm(model)  # our gfx950 NHWC BN kernels
         model = model.to(memory_format=torch.channels_last)
         torch.backends.cudnn.benchmark = True  # MIOpen algo search in warmup
     synchronize_parameters(model)

@@ -22,6 +22,7 @@ sources = [
     os.path.join(CSRC, "ipc_ps.hip"),
     os.path.join(CSRC, "kernels", "reduce.hip"),
     os.path.join(CSRC, "kernels", "pack.hip"),
+    os.path.join(CSRC, "kernels", "batchnorm.hip"),
 ]
 
 extra_cxx = ["-O3", "-std=c++17", "-Wall"]

@@ -20,6 +20,7 @@ namespace tma {
 
 void register_gpu(py::module_& m);     // defined in rccl_engine.hip
 void register_ipc_ps(py::module_& m);  // defined in ipc_ps.hip
+void register_batchnorm(py::module_& m);  // defined in kernels/batchnorm.hip
 
 namespace {
 
@@ -190,6 +191,7 @@ PYBIND11_MODULE(_C, m) {
   // GPU engine + gfx950 kernels + xGMI IPC parameter server.
   register_gpu(m);
   register_ipc_ps(m);
+  register_batchnorm(m);
 }
 
 }  // namespace tma
