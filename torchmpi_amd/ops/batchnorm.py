@@ -80,6 +80,7 @@ class FusedBatchNorm2d(nn.Module):
     @classmethod
     def from_batchnorm(cls, bn: nn.BatchNorm2d) -> "FusedBatchNorm2d":
         m = cls(bn.num_features, bn.eps, bn.momentum if bn.momentum else 0.1)
+        m = m.to(bn.weight.device)
         with torch.no_grad():
             m.weight.copy_(bn.weight.float())
             m.bias.copy_(bn.bias.float())
