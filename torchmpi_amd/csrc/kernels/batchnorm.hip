@@ -52,7 +52,70 @@ __device__ __forceinline__ __hip_bfloat16 stf<__hip_bfloat16>(float v) {
 }
 
 // ---- forward stats: per-channel sum / sumsq ------------------------------
+//
+// Vectorized form (VEC = 8 bf16 / 4 f32 channels = 16 B per lane): a block of
+// 256 threads is laid out as (c8_per_block channel-groups x rows); because
+// c8_per_block divides 256 and nC8 * 16 B = one row's bytes, a wave's 64
+// lanes cover contiguous memory for ANY power-of-two C — full-rate loads.
+// Per-thread fp32 accumulators (VEC of them), block LDS reduction over the
+// row copies, one atomicAdd per (block x channel).
 
+template <typename T, int VEC>
+__global__ __launch_bounds__(256) void bn_stats_vec_kernel(
+    const T* __restrict__ x, float* __restrict__ sum, float* __restrict__ sumsq,
+    size_t M, int nC8, int c8_per_block, int row_blocks) {
+  const int c_blocks = nC8 / c8_per_block;
+  const int cb = blockIdx.x % c_blocks;
+  const int rb = blockIdx.x / c_blocks;
+  const int rows_per_iter = 256 / c8_per_block;
+  const int c8l = threadIdx.x % c8_per_block;          // channel-group in block
+  const int rsub = threadIdx.x / c8_per_block;         // row copy
+  const int c8 = cb * c8_per_block + c8l;
+  float acc[VEC], accsq[VEC];
+#pragma unroll
+  for (int e = 0; e < VEC; ++e) acc[e] = accsq[e] = 0.f;
+  const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+  const T* base = x + (size_t)c8 * VEC;
+  for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+    T v[VEC];
+    *reinterpret_cast<uint4*>(v) =
+        *reinterpret_cast<const uint4*>(base + r * (size_t)nC8 * VEC);
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      float f = ldf(v[e]);
+      acc[e] += f;
+      accsq[e] += f * f;
+    }
+  }
+  // LDS reduction across the row copies of each channel-group.
+  __shared__ float ls[256 * VEC];
+#pragma unroll
+  for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = acc[e];
+  __syncthreads();
+  if (rsub == 0) {
+    for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+      for (int e = 0; e < VEC; ++e)
+        acc[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = accsq[e];
+  __syncthreads();
+  if (rsub == 0) {
+    for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+      for (int e = 0; e < VEC; ++e)
+        accsq[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      atomicAdd(&sum[c8 * VEC + e], acc[e]);
+      atomicAdd(&sumsq[c8 * VEC + e], accsq[e]);
+    }
+  }
+}
+
+// Scalar fallback for C not divisible by the vector width / 256 layout.
 template <typename T>
 __global__ __launch_bounds__(kLanes* kRowsPB) void bn_stats_kernel(
     const T* __restrict__ x, float* __restrict__ sum, float* __restrict__ sumsq,
@@ -150,6 +213,70 @@ __global__ __launch_bounds__(256) void bn_fwd_apply_scalar(
 
 // ---- backward stats: sum(dy), sum(dy * xhat) -----------------------------
 
+template <typename T, int VEC>
+__global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    float* __restrict__ sdy, float* __restrict__ sdyx, size_t M, int nC8,
+    int c8_per_block, int row_blocks) {
+  const int c_blocks = nC8 / c8_per_block;
+  const int cb = blockIdx.x % c_blocks;
+  const int rb = blockIdx.x / c_blocks;
+  const int rows_per_iter = 256 / c8_per_block;
+  const int c8l = threadIdx.x % c8_per_block;
+  const int rsub = threadIdx.x / c8_per_block;
+  const int c8 = cb * c8_per_block + c8l;
+  float mu[VEC], is[VEC], acc[VEC], accx[VEC];
+#pragma unroll
+  for (int e = 0; e < VEC; ++e) {
+    mu[e] = mean[c8 * VEC + e];
+    is[e] = invstd[c8 * VEC + e];
+    acc[e] = accx[e] = 0.f;
+  }
+  const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+  const T* xb = x + (size_t)c8 * VEC;
+  const T* gb = dy + (size_t)c8 * VEC;
+  for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+    T xv[VEC], gv[VEC];
+    *reinterpret_cast<uint4*>(xv) =
+        *reinterpret_cast<const uint4*>(xb + r * (size_t)nC8 * VEC);
+    *reinterpret_cast<uint4*>(gv) =
+        *reinterpret_cast<const uint4*>(gb + r * (size_t)nC8 * VEC);
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      float g = ldf(gv[e]);
+      float xh = (ldf(xv[e]) - mu[e]) * is[e];
+      acc[e] += g;
+      accx[e] += g * xh;
+    }
+  }
+  __shared__ float ls[256 * VEC];
+#pragma unroll
+  for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = acc[e];
+  __syncthreads();
+  if (rsub == 0) {
+    for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+      for (int e = 0; e < VEC; ++e)
+        acc[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = accx[e];
+  __syncthreads();
+  if (rsub == 0) {
+    for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+      for (int e = 0; e < VEC; ++e)
+        accx[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      atomicAdd(&sdy[c8 * VEC + e], acc[e]);
+      atomicAdd(&sdyx[c8 * VEC + e], accx[e]);
+    }
+  }
+}
+
 template <typename T>
 __global__ __launch_bounds__(kLanes* kRowsPB) void bn_bwd_stats_kernel(
     const T* __restrict__ x, const T* __restrict__ dy,
@@ -233,6 +360,34 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_scalar(
   }
 }
 
+// Vector-stats geometry: valid when the channel-group count divides the
+// block cleanly (all power-of-two C, the ResNet/conv-net case).
+struct VecStatsGeom {
+  bool ok = false;
+  int c8_per_block = 0, c_blocks = 0, row_blocks = 0, grid = 0;
+};
+
+VecStatsGeom vec_stats_geom(int C, int VEC, size_t M) {
+  VecStatsGeom g;
+  if (C % VEC != 0) return g;
+  int nC8 = C / VEC;
+  int c8pb;
+  if (nC8 <= 256 && 256 % nC8 == 0)
+    c8pb = nC8;
+  else if (nC8 % 256 == 0)
+    c8pb = 256;
+  else
+    return g;
+  g.c8_per_block = c8pb;
+  g.c_blocks = nC8 / c8pb;
+  int rows_per_iter = 256 / c8pb;
+  int max_rb = (int)((M + rows_per_iter - 1) / rows_per_iter);
+  g.row_blocks = std::min(std::max(2048 / g.c_blocks, 1), std::max(max_rb, 1));
+  g.grid = g.c_blocks * g.row_blocks;
+  g.ok = true;
+  return g;
+}
+
 struct BnGeom {
   size_t M;
   int C, c_blocks, row_blocks;
@@ -278,7 +433,20 @@ std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor b
     at::Tensor sumsq = at::zeros({g.C}, fopt);
     mean = at::empty({g.C}, fopt);
     invstd = at::empty({g.C}, fopt);
-    if (x.scalar_type() == at::kBFloat16)
+    const int vecw = x.scalar_type() == at::kBFloat16 ? 8 : 4;
+    VecStatsGeom vg = vec_stats_geom(g.C, vecw, g.M);
+    if (vg.ok && x.scalar_type() == at::kBFloat16)
+      hipLaunchKernelGGL((bn_stats_vec_kernel<__hip_bfloat16, 8>), dim3(vg.grid),
+                         dim3(256), 0, stream,
+                         (const __hip_bfloat16*)x.data_ptr(),
+                         sum.data_ptr<float>(), sumsq.data_ptr<float>(), g.M,
+                         g.C / 8, vg.c8_per_block, vg.row_blocks);
+    else if (vg.ok)
+      hipLaunchKernelGGL((bn_stats_vec_kernel<float, 4>), dim3(vg.grid),
+                         dim3(256), 0, stream, x.data_ptr<float>(),
+                         sum.data_ptr<float>(), sumsq.data_ptr<float>(), g.M,
+                         g.C / 4, vg.c8_per_block, vg.row_blocks);
+    else if (x.scalar_type() == at::kBFloat16)
       hipLaunchKernelGGL(bn_stats_kernel<__hip_bfloat16>, g.stats_grid,
                          g.stats_block, 0, stream,
                          (const __hip_bfloat16*)x.data_ptr(),
@@ -347,7 +515,24 @@ std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weig
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
   at::Tensor sdy = at::zeros({g.C}, fopt);
   at::Tensor sdyx = at::zeros({g.C}, fopt);
-  if (x.scalar_type() == at::kBFloat16)
+  const int vecw = x.scalar_type() == at::kBFloat16 ? 8 : 4;
+  VecStatsGeom vg = vec_stats_geom(g.C, vecw, g.M);
+  if (vg.ok && x.scalar_type() == at::kBFloat16)
+    hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<__hip_bfloat16, 8>),
+                       dim3(vg.grid), dim3(256), 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (const __hip_bfloat16*)dy.data_ptr(),
+                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(), g.M,
+                       g.C / 8, vg.c8_per_block, vg.row_blocks);
+  else if (vg.ok)
+    hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<float, 4>), dim3(vg.grid),
+                       dim3(256), 0, stream, x.data_ptr<float>(),
+                       dy.data_ptr<float>(), mean.data_ptr<float>(),
+                       invstd.data_ptr<float>(), sdy.data_ptr<float>(),
+                       sdyx.data_ptr<float>(), g.M, g.C / 4, vg.c8_per_block,
+                       vg.row_blocks);
+  else if (x.scalar_type() == at::kBFloat16)
     hipLaunchKernelGGL(bn_bwd_stats_kernel<__hip_bfloat16>, g.stats_grid,
                        g.stats_block, 0, stream,
                        (const __hip_bfloat16*)x.data_ptr(),
