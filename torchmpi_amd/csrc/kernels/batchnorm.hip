@@ -1,13 +1,20 @@
 // Fused NHWC (channels_last) BatchNorm for gfx950 — training fwd + bwd.
 //
 // Motivation (profiles/resnet50_bf16_kernel_stats.md): MIOpen's spatial
-// BatchNorm kernels are ~35% of the ResNet-50 bf16 step on MI355X. These
-// kernels exploit the channels-last layout directly:
-//   - stats/reduction passes: lanes own CONSECUTIVE CHANNELS (64 channels =
-//     one wave = one coalesced 128 B bf16 load per row), fp32 accumulation,
-//     one global atomicAdd per (block x channel),
-//   - apply passes: 16 B/lane vectorized over channels (C % 8 == 0 for
-//     bf16; scalar fallback otherwise).
+// BatchNorm kernels are ~35% of the ResNet-50 bf16 step on MI355X.
+//
+// Design (v2, after ISA inspection of v1 — see profiles/):
+//   - One thread owns a FIXED group of VEC consecutive channels (VEC = 8
+//     bf16 / 4 f32 = 16 B) and loops over rows. Because the channel-group
+//     count divides the 256-thread block and nC8 * 16 B = one row's bytes,
+//     a wave's lanes always cover contiguous memory — full-rate dwordx4
+//     streams for any power-of-two C.
+//   - Fixed channels mean per-channel parameters (mean/invstd/weight/bias/
+//     sums) are loaded ONCE into registers — v1 issued 32 scalar global
+//     loads per 16 B of data (L1-thrashed, latency-bound, 0.3 TB/s).
+//   - Reductions write per-(row-block) partials (plain stores) reduced by
+//     the finalize kernel — v1's global atomicAdd chains were depth
+//     row_blocks (=2048) and cost a flat ~420 us regardless of size.
 // Stats and affine params are fp32; x/y/dy may be bf16 or f32.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -30,9 +37,6 @@ namespace tma {
 
 namespace {
 
-constexpr int kLanes = 64;   // channels per wave
-constexpr int kRowsPB = 4;   // rows (N*H*W positions) per block iteration
-
 template <typename T>
 __device__ __forceinline__ float ldf(T v);
 template <>
@@ -51,26 +55,46 @@ __device__ __forceinline__ __hip_bfloat16 stf<__hip_bfloat16>(float v) {
   return __float2bfloat16(v);
 }
 
-// ---- forward stats: per-channel sum / sumsq ------------------------------
-//
-// Vectorized form (VEC = 8 bf16 / 4 f32 channels = 16 B per lane): a block of
-// 256 threads is laid out as (c8_per_block channel-groups x rows); because
-// c8_per_block divides 256 and nC8 * 16 B = one row's bytes, a wave's 64
-// lanes cover contiguous memory for ANY power-of-two C — full-rate loads.
-// Per-thread fp32 accumulators (VEC of them), block LDS reduction over the
-// row copies, one atomicAdd per (block x channel).
+// Geometry shared by all vector kernels: block = (c8_per_block channel
+// groups) x (256 / c8_per_block row copies); grid = c_blocks x row_blocks.
+struct VecGeom {
+  bool ok = false;
+  int nC8 = 0, c8_per_block = 0, c_blocks = 0, row_blocks = 0, grid = 0;
+};
+
+VecGeom vec_geom(int C, int VEC, size_t M) {
+  VecGeom g;
+  if (C % VEC != 0) return g;
+  g.nC8 = C / VEC;
+  if (g.nC8 <= 256 && 256 % g.nC8 == 0)
+    g.c8_per_block = g.nC8;
+  else if (g.nC8 % 256 == 0)
+    g.c8_per_block = 256;
+  else
+    return g;
+  g.c_blocks = g.nC8 / g.c8_per_block;
+  int rows_per_iter = 256 / g.c8_per_block;
+  int max_rb = (int)((M + rows_per_iter - 1) / rows_per_iter);
+  g.row_blocks = std::min(std::max(4096 / g.c_blocks, 1), std::max(max_rb, 1));
+  g.grid = g.c_blocks * g.row_blocks;
+  g.ok = true;
+  return g;
+}
+
+// ---- vector stats: per-(row_block, channel) partial sums -----------------
 
 template <typename T, int VEC>
 __global__ __launch_bounds__(256) void bn_stats_vec_kernel(
-    const T* __restrict__ x, float* __restrict__ sum, float* __restrict__ sumsq,
+    const T* __restrict__ x, float* __restrict__ psum, float* __restrict__ psumsq,
     size_t M, int nC8, int c8_per_block, int row_blocks) {
   const int c_blocks = nC8 / c8_per_block;
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
   const int rows_per_iter = 256 / c8_per_block;
-  const int c8l = threadIdx.x % c8_per_block;          // channel-group in block
-  const int rsub = threadIdx.x / c8_per_block;         // row copy
+  const int c8l = (int)threadIdx.x % c8_per_block;
+  const int rsub = (int)threadIdx.x / c8_per_block;
   const int c8 = cb * c8_per_block + c8l;
+  const int C = nC8 * VEC;
   float acc[VEC], accsq[VEC];
 #pragma unroll
   for (int e = 0; e < VEC; ++e) acc[e] = accsq[e] = 0.f;
@@ -79,7 +103,7 @@ __global__ __launch_bounds__(256) void bn_stats_vec_kernel(
   for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
     T v[VEC];
     *reinterpret_cast<uint4*>(v) =
-        *reinterpret_cast<const uint4*>(base + r * (size_t)nC8 * VEC);
+        *reinterpret_cast<const uint4*>(base + r * (size_t)C);
 #pragma unroll
     for (int e = 0; e < VEC; ++e) {
       float f = ldf(v[e]);
@@ -87,7 +111,6 @@ __global__ __launch_bounds__(256) void bn_stats_vec_kernel(
       accsq[e] += f * f;
     }
   }
-  // LDS reduction across the row copies of each channel-group.
   __shared__ float ls[256 * VEC];
 #pragma unroll
   for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = acc[e];
@@ -107,62 +130,72 @@ __global__ __launch_bounds__(256) void bn_stats_vec_kernel(
 #pragma unroll
       for (int e = 0; e < VEC; ++e)
         accsq[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+    float* ps = psum + (size_t)rb * C + c8 * VEC;
+    float* pq = psumsq + (size_t)rb * C + c8 * VEC;
 #pragma unroll
-    for (int e = 0; e < VEC; ++e) {
-      atomicAdd(&sum[c8 * VEC + e], acc[e]);
-      atomicAdd(&sumsq[c8 * VEC + e], accsq[e]);
+    for (int e = 0; e < VEC; e += 4) {
+      *reinterpret_cast<float4*>(ps + e) =
+          make_float4(acc[e], acc[e + 1], acc[e + 2], acc[e + 3]);
+      *reinterpret_cast<float4*>(pq + e) =
+          make_float4(accsq[e], accsq[e + 1], accsq[e + 2], accsq[e + 3]);
     }
   }
 }
 
-// Scalar fallback for C not divisible by the vector width / 256 layout.
+// Scalar fallback (odd C): atomics into sum/sumsq directly — cold path.
 template <typename T>
-__global__ __launch_bounds__(kLanes* kRowsPB) void bn_stats_kernel(
+__global__ __launch_bounds__(256) void bn_stats_scalar_kernel(
     const T* __restrict__ x, float* __restrict__ sum, float* __restrict__ sumsq,
     size_t M, int C, int c_blocks, int row_blocks) {
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
-  const int c = cb * kLanes + threadIdx.x;
+  const int lane = (int)threadIdx.x % 64;
+  const int rsub = (int)threadIdx.x / 64;
+  const int c = cb * 64 + lane;
   float s = 0.f, ss = 0.f;
   if (c < C) {
-    const size_t row_stride = (size_t)row_blocks * kRowsPB;
-    for (size_t r = (size_t)rb * kRowsPB + threadIdx.y; r < M; r += row_stride) {
+    const size_t row_stride = (size_t)row_blocks * 4;
+    for (size_t r = (size_t)rb * 4 + rsub; r < M; r += row_stride) {
       float v = ldf(x[r * C + c]);
       s += v;
       ss += v * v;
     }
   }
-  // reduce the kRowsPB partials per channel through LDS
-  __shared__ float ls[kRowsPB][kLanes];
-  __shared__ float lss[kRowsPB][kLanes];
-  ls[threadIdx.y][threadIdx.x] = s;
-  lss[threadIdx.y][threadIdx.x] = ss;
+  __shared__ float ls[4][64];
+  __shared__ float lss[4][64];
+  ls[rsub][lane] = s;
+  lss[rsub][lane] = ss;
   __syncthreads();
-  if (threadIdx.y == 0 && c < C) {
+  if (rsub == 0 && c < C) {
     float ts = 0.f, tss = 0.f;
 #pragma unroll
-    for (int w = 0; w < kRowsPB; ++w) {
-      ts += ls[w][threadIdx.x];
-      tss += lss[w][threadIdx.x];
+    for (int w = 0; w < 4; ++w) {
+      ts += ls[w][lane];
+      tss += lss[w][lane];
     }
     atomicAdd(&sum[c], ts);
     atomicAdd(&sumsq[c], tss);
   }
 }
 
-// ---- finalize: mean/invstd + running stats -------------------------------
+// ---- finalize: reduce partials + mean/invstd + running stats -------------
 
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq,
-                                   float* __restrict__ mean,
+__global__ void bn_finalize_kernel(const float* __restrict__ psum,
+                                   const float* __restrict__ psumsq,
+                                   int row_blocks, float* __restrict__ mean,
                                    float* __restrict__ invstd,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var, float momentum,
                                    float eps, float M, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float m = sum[c] / M;
-  float var = fmaxf(sumsq[c] / M - m * m, 0.f);
+  float s = 0.f, ss = 0.f;
+  for (int rb = 0; rb < row_blocks; ++rb) {
+    s += psum[(size_t)rb * C + c];
+    ss += psumsq[(size_t)rb * C + c];
+  }
+  float m = s / M;
+  float var = fmaxf(ss / M - m * m, 0.f);
   mean[c] = m;
   invstd[c] = rsqrtf(var + eps);
   if (running_mean) {
@@ -172,28 +205,54 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
   }
 }
 
-// ---- forward apply -------------------------------------------------------
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ psdy,
+                                       const float* __restrict__ psdyx,
+                                       int row_blocks, float* __restrict__ sdy,
+                                       float* __restrict__ sdyx, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.f, b = 0.f;
+  for (int rb = 0; rb < row_blocks; ++rb) {
+    a += psdy[(size_t)rb * C + c];
+    b += psdyx[(size_t)rb * C + c];
+  }
+  sdy[c] = a;
+  sdyx[c] = b;
+}
+
+// ---- forward apply (fixed channels per thread, params in registers) ------
 
 template <typename T, int VEC>
-__global__ __launch_bounds__(256) void bn_fwd_apply_kernel(
+__global__ __launch_bounds__(256) void bn_fwd_apply_vec(
     const T* __restrict__ x, T* __restrict__ y, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ weight,
-    const float* __restrict__ bias, size_t total, int C) {
-  const size_t nvec = total / VEC;
-  const size_t stride = (size_t)gridDim.x * blockDim.x;
-  for (size_t v = (size_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec;
-       v += stride) {
-    size_t idx = v * VEC;
-    int c0 = (int)(idx % C);
-    T in[VEC], out[VEC];
-    *reinterpret_cast<uint4*>(in) = reinterpret_cast<const uint4*>(x)[v];
+    const float* __restrict__ bias, size_t M, int nC8, int c8_per_block,
+    int row_blocks) {
+  const int c_blocks = nC8 / c8_per_block;
+  const int cb = blockIdx.x % c_blocks;
+  const int rb = blockIdx.x / c_blocks;
+  const int rows_per_iter = 256 / c8_per_block;
+  const int c8l = (int)threadIdx.x % c8_per_block;
+  const int rsub = (int)threadIdx.x / c8_per_block;
+  const int c8 = cb * c8_per_block + c8l;
+  const int C = nC8 * VEC;
+  float sc[VEC], sh[VEC];
 #pragma unroll
-    for (int e = 0; e < VEC; ++e) {
-      int c = c0 + e;
-      float xh = (ldf(in[e]) - mean[c]) * invstd[c];
-      out[e] = stf<T>(xh * weight[c] + bias[c]);
-    }
-    reinterpret_cast<uint4*>(y)[v] = *reinterpret_cast<uint4*>(out);
+  for (int e = 0; e < VEC; ++e) {
+    int c = c8 * VEC + e;
+    sc[e] = weight[c] * invstd[c];  // y = sc * x + sh
+    sh[e] = bias[c] - mean[c] * sc[e];
+  }
+  const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+  const T* xb = x + (size_t)c8 * VEC;
+  T* yb = y + (size_t)c8 * VEC;
+  for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+    T v[VEC], o[VEC];
+    *reinterpret_cast<uint4*>(v) =
+        *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) o[e] = stf<T>(ldf(v[e]) * sc[e] + sh[e]);
+    *reinterpret_cast<uint4*>(yb + r * (size_t)C) = *reinterpret_cast<uint4*>(o);
   }
 }
 
@@ -211,21 +270,22 @@ __global__ __launch_bounds__(256) void bn_fwd_apply_scalar(
   }
 }
 
-// ---- backward stats: sum(dy), sum(dy * xhat) -----------------------------
+// ---- backward stats ------------------------------------------------------
 
 template <typename T, int VEC>
 __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
     const T* __restrict__ x, const T* __restrict__ dy,
     const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ sdy, float* __restrict__ sdyx, size_t M, int nC8,
+    float* __restrict__ psdy, float* __restrict__ psdyx, size_t M, int nC8,
     int c8_per_block, int row_blocks) {
   const int c_blocks = nC8 / c8_per_block;
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
   const int rows_per_iter = 256 / c8_per_block;
-  const int c8l = threadIdx.x % c8_per_block;
-  const int rsub = threadIdx.x / c8_per_block;
+  const int c8l = (int)threadIdx.x % c8_per_block;
+  const int rsub = (int)threadIdx.x / c8_per_block;
   const int c8 = cb * c8_per_block + c8l;
+  const int C = nC8 * VEC;
   float mu[VEC], is[VEC], acc[VEC], accx[VEC];
 #pragma unroll
   for (int e = 0; e < VEC; ++e) {
@@ -239,9 +299,9 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
   for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
     T xv[VEC], gv[VEC];
     *reinterpret_cast<uint4*>(xv) =
-        *reinterpret_cast<const uint4*>(xb + r * (size_t)nC8 * VEC);
+        *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
     *reinterpret_cast<uint4*>(gv) =
-        *reinterpret_cast<const uint4*>(gb + r * (size_t)nC8 * VEC);
+        *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
 #pragma unroll
     for (int e = 0; e < VEC; ++e) {
       float g = ldf(gv[e]);
@@ -269,45 +329,51 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
 #pragma unroll
       for (int e = 0; e < VEC; ++e)
         accx[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+    float* pa = psdy + (size_t)rb * C + c8 * VEC;
+    float* pb = psdyx + (size_t)rb * C + c8 * VEC;
 #pragma unroll
-    for (int e = 0; e < VEC; ++e) {
-      atomicAdd(&sdy[c8 * VEC + e], acc[e]);
-      atomicAdd(&sdyx[c8 * VEC + e], accx[e]);
+    for (int e = 0; e < VEC; e += 4) {
+      *reinterpret_cast<float4*>(pa + e) =
+          make_float4(acc[e], acc[e + 1], acc[e + 2], acc[e + 3]);
+      *reinterpret_cast<float4*>(pb + e) =
+          make_float4(accx[e], accx[e + 1], accx[e + 2], accx[e + 3]);
     }
   }
 }
 
 template <typename T>
-__global__ __launch_bounds__(kLanes* kRowsPB) void bn_bwd_stats_kernel(
+__global__ __launch_bounds__(256) void bn_bwd_stats_scalar_kernel(
     const T* __restrict__ x, const T* __restrict__ dy,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     float* __restrict__ sdy, float* __restrict__ sdyx, size_t M, int C,
     int c_blocks, int row_blocks) {
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
-  const int c = cb * kLanes + threadIdx.x;
+  const int lane = (int)threadIdx.x % 64;
+  const int rsub = (int)threadIdx.x / 64;
+  const int c = cb * 64 + lane;
   float s = 0.f, sx = 0.f;
   if (c < C) {
     const float mu = mean[c], is = invstd[c];
-    const size_t row_stride = (size_t)row_blocks * kRowsPB;
-    for (size_t r = (size_t)rb * kRowsPB + threadIdx.y; r < M; r += row_stride) {
+    const size_t row_stride = (size_t)row_blocks * 4;
+    for (size_t r = (size_t)rb * 4 + rsub; r < M; r += row_stride) {
       float g = ldf(dy[r * C + c]);
       float xh = (ldf(x[r * C + c]) - mu) * is;
       s += g;
       sx += g * xh;
     }
   }
-  __shared__ float ls[kRowsPB][kLanes];
-  __shared__ float lsx[kRowsPB][kLanes];
-  ls[threadIdx.y][threadIdx.x] = s;
-  lsx[threadIdx.y][threadIdx.x] = sx;
+  __shared__ float ls[4][64];
+  __shared__ float lsx[4][64];
+  ls[rsub][lane] = s;
+  lsx[rsub][lane] = sx;
   __syncthreads();
-  if (threadIdx.y == 0 && c < C) {
+  if (rsub == 0 && c < C) {
     float ts = 0.f, tsx = 0.f;
 #pragma unroll
-    for (int w = 0; w < kRowsPB; ++w) {
-      ts += ls[w][threadIdx.x];
-      tsx += lsx[w][threadIdx.x];
+    for (int w = 0; w < 4; ++w) {
+      ts += ls[w][lane];
+      tsx += lsx[w][lane];
     }
     atomicAdd(&sdy[c], ts);
     atomicAdd(&sdyx[c], tsx);
@@ -317,29 +383,45 @@ __global__ __launch_bounds__(kLanes* kRowsPB) void bn_bwd_stats_kernel(
 // ---- backward apply ------------------------------------------------------
 
 template <typename T, int VEC>
-__global__ __launch_bounds__(256) void bn_bwd_apply_kernel(
+__global__ __launch_bounds__(256) void bn_bwd_apply_vec(
     const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ sdy,
-    const float* __restrict__ sdyx, float invM, size_t total, int C) {
-  const size_t nvec = total / VEC;
-  const size_t stride = (size_t)gridDim.x * blockDim.x;
-  for (size_t v = (size_t)blockIdx.x * blockDim.x + threadIdx.x; v < nvec;
-       v += stride) {
-    size_t idx = v * VEC;
-    int c0 = (int)(idx % C);
-    T xin[VEC], gin[VEC], out[VEC];
-    *reinterpret_cast<uint4*>(xin) = reinterpret_cast<const uint4*>(x)[v];
-    *reinterpret_cast<uint4*>(gin) = reinterpret_cast<const uint4*>(dy)[v];
+    const float* __restrict__ sdyx, float invM, size_t M, int nC8,
+    int c8_per_block, int row_blocks) {
+  const int c_blocks = nC8 / c8_per_block;
+  const int cb = blockIdx.x % c_blocks;
+  const int rb = blockIdx.x / c_blocks;
+  const int rows_per_iter = 256 / c8_per_block;
+  const int c8l = (int)threadIdx.x % c8_per_block;
+  const int rsub = (int)threadIdx.x / c8_per_block;
+  const int c8 = cb * c8_per_block + c8l;
+  const int C = nC8 * VEC;
+  // dx = w*is*(dy - sdy/M - xhat*sdyx/M) = A*dy + B*x + D
+  float A[VEC], B[VEC], D[VEC];
 #pragma unroll
-    for (int e = 0; e < VEC; ++e) {
-      int c = c0 + e;
-      float xh = (ldf(xin[e]) - mean[c]) * invstd[c];
-      float g = ldf(gin[e]);
-      float t = g - sdy[c] * invM - xh * sdyx[c] * invM;
-      out[e] = stf<T>(weight[c] * invstd[c] * t);
-    }
-    reinterpret_cast<uint4*>(dx)[v] = *reinterpret_cast<uint4*>(out);
+  for (int e = 0; e < VEC; ++e) {
+    int c = c8 * VEC + e;
+    float a = weight[c] * invstd[c];
+    float b = -a * invstd[c] * sdyx[c] * invM;
+    A[e] = a;
+    B[e] = b;
+    D[e] = -a * sdy[c] * invM - b * mean[c];
+  }
+  const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+  const T* xb = x + (size_t)c8 * VEC;
+  const T* gb = dy + (size_t)c8 * VEC;
+  T* ob = dx + (size_t)c8 * VEC;
+  for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+    T xv[VEC], gv[VEC], o[VEC];
+    *reinterpret_cast<uint4*>(xv) =
+        *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
+    *reinterpret_cast<uint4*>(gv) =
+        *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
+#pragma unroll
+    for (int e = 0; e < VEC; ++e)
+      o[e] = stf<T>(A[e] * ldf(gv[e]) + B[e] * ldf(xv[e]) + D[e]);
+    *reinterpret_cast<uint4*>(ob + r * (size_t)C) = *reinterpret_cast<uint4*>(o);
   }
 }
 
@@ -360,61 +442,17 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_scalar(
   }
 }
 
-// Vector-stats geometry: valid when the channel-group count divides the
-// block cleanly (all power-of-two C, the ResNet/conv-net case).
-struct VecStatsGeom {
-  bool ok = false;
-  int c8_per_block = 0, c_blocks = 0, row_blocks = 0, grid = 0;
-};
-
-VecStatsGeom vec_stats_geom(int C, int VEC, size_t M) {
-  VecStatsGeom g;
-  if (C % VEC != 0) return g;
-  int nC8 = C / VEC;
-  int c8pb;
-  if (nC8 <= 256 && 256 % nC8 == 0)
-    c8pb = nC8;
-  else if (nC8 % 256 == 0)
-    c8pb = 256;
-  else
-    return g;
-  g.c8_per_block = c8pb;
-  g.c_blocks = nC8 / c8pb;
-  int rows_per_iter = 256 / c8pb;
-  int max_rb = (int)((M + rows_per_iter - 1) / rows_per_iter);
-  g.row_blocks = std::min(std::max(2048 / g.c_blocks, 1), std::max(max_rb, 1));
-  g.grid = g.c_blocks * g.row_blocks;
-  g.ok = true;
-  return g;
-}
-
-struct BnGeom {
-  size_t M;
-  int C, c_blocks, row_blocks;
-  dim3 stats_grid, stats_block;
-  int apply_grid;
-};
-
-BnGeom bn_geom(const at::Tensor& x) {
-  BnGeom g;
-  g.C = (int)x.size(1);
-  g.M = (size_t)x.numel() / g.C;
-  g.c_blocks = (g.C + kLanes - 1) / kLanes;
-  int max_row_blocks = (int)((g.M + kRowsPB - 1) / kRowsPB);
-  g.row_blocks = std::min(std::max(2048 / g.c_blocks, 1), max_row_blocks);
-  g.stats_grid = dim3(g.c_blocks * g.row_blocks);
-  g.stats_block = dim3(kLanes, kRowsPB);
-  size_t work = x.numel() / 8 + 1;
-  g.apply_grid = (int)std::min<size_t>((work + 255) / 256, 2048);
-  return g;
-}
-
 void check_bn_input(const at::Tensor& x) {
   TMA_CHECK(x.is_cuda() && x.dim() == 4, "bn: 4D CUDA tensor required");
   TMA_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
             "bn: channels_last layout required");
   TMA_CHECK(x.scalar_type() == at::kBFloat16 || x.scalar_type() == at::kFloat,
             "bn: bf16 or f32 input");
+}
+
+int scalar_grid(size_t work) {
+  size_t blocks = (work + 255) / 256;
+  return (int)std::min<size_t>(blocks ? blocks : 1, 2048);
 }
 
 }  // namespace
@@ -425,82 +463,100 @@ std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor b
                                    bool training, double momentum, double eps) {
   check_bn_input(x);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  BnGeom g = bn_geom(x);
+  const int C = (int)x.size(1);
+  const size_t M = (size_t)x.numel() / C;
+  const bool bf16 = x.scalar_type() == at::kBFloat16;
+  const int vecw = bf16 ? 8 : 4;
+  VecGeom vg = vec_geom(C, vecw, M);
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
   at::Tensor mean, invstd;
+
   if (training) {
-    at::Tensor sum = at::zeros({g.C}, fopt);
-    at::Tensor sumsq = at::zeros({g.C}, fopt);
-    mean = at::empty({g.C}, fopt);
-    invstd = at::empty({g.C}, fopt);
-    const int vecw = x.scalar_type() == at::kBFloat16 ? 8 : 4;
-    VecStatsGeom vg = vec_stats_geom(g.C, vecw, g.M);
-    if (vg.ok && x.scalar_type() == at::kBFloat16)
-      hipLaunchKernelGGL((bn_stats_vec_kernel<__hip_bfloat16, 8>), dim3(vg.grid),
-                         dim3(256), 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         sum.data_ptr<float>(), sumsq.data_ptr<float>(), g.M,
-                         g.C / 8, vg.c8_per_block, vg.row_blocks);
-    else if (vg.ok)
-      hipLaunchKernelGGL((bn_stats_vec_kernel<float, 4>), dim3(vg.grid),
-                         dim3(256), 0, stream, x.data_ptr<float>(),
-                         sum.data_ptr<float>(), sumsq.data_ptr<float>(), g.M,
-                         g.C / 4, vg.c8_per_block, vg.row_blocks);
-    else if (x.scalar_type() == at::kBFloat16)
-      hipLaunchKernelGGL(bn_stats_kernel<__hip_bfloat16>, g.stats_grid,
-                         g.stats_block, 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         sum.data_ptr<float>(), sumsq.data_ptr<float>(), g.M, g.C,
-                         g.c_blocks, g.row_blocks);
-    else
-      hipLaunchKernelGGL(bn_stats_kernel<float>, g.stats_grid, g.stats_block, 0,
-                         stream, x.data_ptr<float>(), sum.data_ptr<float>(),
-                         sumsq.data_ptr<float>(), g.M, g.C, g.c_blocks,
-                         g.row_blocks);
-    hipLaunchKernelGGL(bn_finalize_kernel, dim3((g.C + 255) / 256), dim3(256), 0,
-                       stream, sum.data_ptr<float>(), sumsq.data_ptr<float>(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       running_mean.defined() ? running_mean.data_ptr<float>()
-                                              : nullptr,
-                       running_var.defined() ? running_var.data_ptr<float>()
-                                             : nullptr,
-                       (float)momentum, (float)eps, (float)g.M, g.C);
+    mean = at::empty({C}, fopt);
+    invstd = at::empty({C}, fopt);
+    if (vg.ok) {
+      at::Tensor psum = at::empty({(int64_t)vg.row_blocks * C}, fopt);
+      at::Tensor psumsq = at::empty({(int64_t)vg.row_blocks * C}, fopt);
+      if (bf16)
+        hipLaunchKernelGGL((bn_stats_vec_kernel<__hip_bfloat16, 8>),
+                           dim3(vg.grid), dim3(256), 0, stream,
+                           (const __hip_bfloat16*)x.data_ptr(),
+                           psum.data_ptr<float>(), psumsq.data_ptr<float>(), M,
+                           vg.nC8, vg.c8_per_block, vg.row_blocks);
+      else
+        hipLaunchKernelGGL((bn_stats_vec_kernel<float, 4>), dim3(vg.grid),
+                           dim3(256), 0, stream, x.data_ptr<float>(),
+                           psum.data_ptr<float>(), psumsq.data_ptr<float>(), M,
+                           vg.nC8, vg.c8_per_block, vg.row_blocks);
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                         stream, psum.data_ptr<float>(), psumsq.data_ptr<float>(),
+                         vg.row_blocks, mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(),
+                         running_mean.defined() ? running_mean.data_ptr<float>()
+                                                : nullptr,
+                         running_var.defined() ? running_var.data_ptr<float>()
+                                               : nullptr,
+                         (float)momentum, (float)eps, (float)M, C);
+    } else {
+      at::Tensor sum = at::zeros({C}, fopt);
+      at::Tensor sumsq = at::zeros({C}, fopt);
+      int c_blocks = (C + 63) / 64;
+      int row_blocks = std::max(1, 512 / c_blocks);
+      dim3 grid(c_blocks * row_blocks), block(256);
+      if (bf16)
+        hipLaunchKernelGGL(bn_stats_scalar_kernel<__hip_bfloat16>, grid, block, 0,
+                           stream, (const __hip_bfloat16*)x.data_ptr(),
+                           sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C,
+                           c_blocks, row_blocks);
+      else
+        hipLaunchKernelGGL(bn_stats_scalar_kernel<float>, grid, block, 0, stream,
+                           x.data_ptr<float>(), sum.data_ptr<float>(),
+                           sumsq.data_ptr<float>(), M, C, c_blocks, row_blocks);
+      hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                         stream, sum.data_ptr<float>(), sumsq.data_ptr<float>(),
+                         1, mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         running_mean.defined() ? running_mean.data_ptr<float>()
+                                                : nullptr,
+                         running_var.defined() ? running_var.data_ptr<float>()
+                                               : nullptr,
+                         (float)momentum, (float)eps, (float)M, C);
+    }
   } else {
-    mean = running_mean;
+    mean = running_mean.contiguous();
     invstd = (running_var + eps).rsqrt();
   }
+
   at::Tensor y = at::empty_like(x);
-  size_t total = (size_t)x.numel();
-  bool vec_ok = (g.C % 8 == 0) && x.scalar_type() == at::kBFloat16;
-  bool vec4_ok = (g.C % 4 == 0) && x.scalar_type() == at::kFloat;
-  if (x.scalar_type() == at::kBFloat16) {
-    if (vec_ok)
-      hipLaunchKernelGGL((bn_fwd_apply_kernel<__hip_bfloat16, 8>),
-                         dim3(g.apply_grid), dim3(256), 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         (__hip_bfloat16*)y.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         bias.data_ptr<float>(), total, g.C);
-    else
-      hipLaunchKernelGGL(bn_fwd_apply_scalar<__hip_bfloat16>, dim3(g.apply_grid),
+  if (vg.ok) {
+    if (bf16)
+      hipLaunchKernelGGL((bn_fwd_apply_vec<__hip_bfloat16, 8>), dim3(vg.grid),
                          dim3(256), 0, stream,
                          (const __hip_bfloat16*)x.data_ptr(),
                          (__hip_bfloat16*)y.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         bias.data_ptr<float>(), total, g.C);
+                         bias.data_ptr<float>(), M, vg.nC8, vg.c8_per_block,
+                         vg.row_blocks);
+    else
+      hipLaunchKernelGGL((bn_fwd_apply_vec<float, 4>), dim3(vg.grid), dim3(256),
+                         0, stream, x.data_ptr<float>(), y.data_ptr<float>(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         weight.data_ptr<float>(), bias.data_ptr<float>(), M,
+                         vg.nC8, vg.c8_per_block, vg.row_blocks);
   } else {
-    if (vec4_ok)
-      hipLaunchKernelGGL((bn_fwd_apply_kernel<float, 4>), dim3(g.apply_grid),
+    size_t total = (size_t)x.numel();
+    if (bf16)
+      hipLaunchKernelGGL(bn_fwd_apply_scalar<__hip_bfloat16>,
+                         dim3(scalar_grid(total)), dim3(256), 0, stream,
+                         (const __hip_bfloat16*)x.data_ptr(),
+                         (__hip_bfloat16*)y.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
+                         bias.data_ptr<float>(), total, C);
+    else
+      hipLaunchKernelGGL(bn_fwd_apply_scalar<float>, dim3(scalar_grid(total)),
                          dim3(256), 0, stream, x.data_ptr<float>(),
                          y.data_ptr<float>(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         bias.data_ptr<float>(), total, g.C);
-    else
-      hipLaunchKernelGGL(bn_fwd_apply_scalar<float>, dim3(g.apply_grid), dim3(256),
-                         0, stream, x.data_ptr<float>(), y.data_ptr<float>(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         weight.data_ptr<float>(), bias.data_ptr<float>(), total,
-                         g.C);
+                         bias.data_ptr<float>(), total, C);
   }
   return {y, mean, invstd};
 }
@@ -511,78 +567,95 @@ std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weig
   check_bn_input(x);
   check_bn_input(dy);
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  BnGeom g = bn_geom(x);
+  const int C = (int)x.size(1);
+  const size_t M = (size_t)x.numel() / C;
+  const bool bf16 = x.scalar_type() == at::kBFloat16;
+  const int vecw = bf16 ? 8 : 4;
+  VecGeom vg = vec_geom(C, vecw, M);
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
-  at::Tensor sdy = at::zeros({g.C}, fopt);
-  at::Tensor sdyx = at::zeros({g.C}, fopt);
-  const int vecw = x.scalar_type() == at::kBFloat16 ? 8 : 4;
-  VecStatsGeom vg = vec_stats_geom(g.C, vecw, g.M);
-  if (vg.ok && x.scalar_type() == at::kBFloat16)
-    hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<__hip_bfloat16, 8>),
-                       dim3(vg.grid), dim3(256), 0, stream,
-                       (const __hip_bfloat16*)x.data_ptr(),
-                       (const __hip_bfloat16*)dy.data_ptr(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(), g.M,
-                       g.C / 8, vg.c8_per_block, vg.row_blocks);
-  else if (vg.ok)
-    hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<float, 4>), dim3(vg.grid),
-                       dim3(256), 0, stream, x.data_ptr<float>(),
-                       dy.data_ptr<float>(), mean.data_ptr<float>(),
-                       invstd.data_ptr<float>(), sdy.data_ptr<float>(),
-                       sdyx.data_ptr<float>(), g.M, g.C / 4, vg.c8_per_block,
-                       vg.row_blocks);
-  else if (x.scalar_type() == at::kBFloat16)
-    hipLaunchKernelGGL(bn_bwd_stats_kernel<__hip_bfloat16>, g.stats_grid,
-                       g.stats_block, 0, stream,
-                       (const __hip_bfloat16*)x.data_ptr(),
-                       (const __hip_bfloat16*)dy.data_ptr(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(), g.M, g.C,
-                       g.c_blocks, g.row_blocks);
-  else
-    hipLaunchKernelGGL(bn_bwd_stats_kernel<float>, g.stats_grid, g.stats_block, 0,
-                       stream, x.data_ptr<float>(), dy.data_ptr<float>(),
-                       mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(), g.M, g.C,
-                       g.c_blocks, g.row_blocks);
-  at::Tensor dx = at::empty_like(x);
-  size_t total = (size_t)x.numel();
-  float invM = 1.f / (float)g.M;
-  if (x.scalar_type() == at::kBFloat16) {
-    if (g.C % 8 == 0)
-      hipLaunchKernelGGL((bn_bwd_apply_kernel<__hip_bfloat16, 8>),
-                         dim3(g.apply_grid), dim3(256), 0, stream,
+  at::Tensor sdy = at::empty({C}, fopt);
+  at::Tensor sdyx = at::empty({C}, fopt);
+
+  if (vg.ok) {
+    at::Tensor psdy = at::empty({(int64_t)vg.row_blocks * C}, fopt);
+    at::Tensor psdyx = at::empty({(int64_t)vg.row_blocks * C}, fopt);
+    if (bf16)
+      hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<__hip_bfloat16, 8>),
+                         dim3(vg.grid), dim3(256), 0, stream,
                          (const __hip_bfloat16*)x.data_ptr(),
                          (const __hip_bfloat16*)dy.data_ptr(),
-                         (__hip_bfloat16*)dx.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM,
-                         total, g.C);
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         psdy.data_ptr<float>(), psdyx.data_ptr<float>(), M,
+                         vg.nC8, vg.c8_per_block, vg.row_blocks);
     else
-      hipLaunchKernelGGL(bn_bwd_apply_scalar<__hip_bfloat16>, dim3(g.apply_grid),
+      hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<float, 4>), dim3(vg.grid),
+                         dim3(256), 0, stream, x.data_ptr<float>(),
+                         dy.data_ptr<float>(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), psdy.data_ptr<float>(),
+                         psdyx.data_ptr<float>(), M, vg.nC8, vg.c8_per_block,
+                         vg.row_blocks);
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, stream, psdy.data_ptr<float>(), psdyx.data_ptr<float>(),
+                       vg.row_blocks, sdy.data_ptr<float>(),
+                       sdyx.data_ptr<float>(), C);
+  } else {
+    sdy.zero_();
+    sdyx.zero_();
+    int c_blocks = (C + 63) / 64;
+    int row_blocks = std::max(1, 512 / c_blocks);
+    dim3 grid(c_blocks * row_blocks), block(256);
+    if (bf16)
+      hipLaunchKernelGGL(bn_bwd_stats_scalar_kernel<__hip_bfloat16>, grid, block,
+                         0, stream, (const __hip_bfloat16*)x.data_ptr(),
+                         (const __hip_bfloat16*)dy.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), M, C,
+                         c_blocks, row_blocks);
+    else
+      hipLaunchKernelGGL(bn_bwd_stats_scalar_kernel<float>, grid, block, 0,
+                         stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), M, C,
+                         c_blocks, row_blocks);
+  }
+
+  at::Tensor dx = at::empty_like(x);
+  float invM = 1.f / (float)M;
+  if (vg.ok) {
+    if (bf16)
+      hipLaunchKernelGGL((bn_bwd_apply_vec<__hip_bfloat16, 8>), dim3(vg.grid),
                          dim3(256), 0, stream,
                          (const __hip_bfloat16*)x.data_ptr(),
                          (const __hip_bfloat16*)dy.data_ptr(),
                          (__hip_bfloat16*)dx.data_ptr(), mean.data_ptr<float>(),
                          invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM,
-                         total, g.C);
+                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM, M,
+                         vg.nC8, vg.c8_per_block, vg.row_blocks);
+    else
+      hipLaunchKernelGGL((bn_bwd_apply_vec<float, 4>), dim3(vg.grid), dim3(256),
+                         0, stream, x.data_ptr<float>(), dy.data_ptr<float>(),
+                         dx.data_ptr<float>(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
+                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM, M,
+                         vg.nC8, vg.c8_per_block, vg.row_blocks);
   } else {
-    if (g.C % 4 == 0)
-      hipLaunchKernelGGL((bn_bwd_apply_kernel<float, 4>), dim3(g.apply_grid),
+    size_t total = (size_t)x.numel();
+    if (bf16)
+      hipLaunchKernelGGL(bn_bwd_apply_scalar<__hip_bfloat16>,
+                         dim3(scalar_grid(total)), dim3(256), 0, stream,
+                         (const __hip_bfloat16*)x.data_ptr(),
+                         (const __hip_bfloat16*)dy.data_ptr(),
+                         (__hip_bfloat16*)dx.data_ptr(), mean.data_ptr<float>(),
+                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
+                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM,
+                         total, C);
+    else
+      hipLaunchKernelGGL(bn_bwd_apply_scalar<float>, dim3(scalar_grid(total)),
                          dim3(256), 0, stream, x.data_ptr<float>(),
                          dy.data_ptr<float>(), dx.data_ptr<float>(),
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          weight.data_ptr<float>(), sdy.data_ptr<float>(),
-                         sdyx.data_ptr<float>(), invM, total, g.C);
-    else
-      hipLaunchKernelGGL(bn_bwd_apply_scalar<float>, dim3(g.apply_grid), dim3(256),
-                         0, stream, x.data_ptr<float>(), dy.data_ptr<float>(),
-                         dx.data_ptr<float>(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM,
-                         total, g.C);
+                         sdyx.data_ptr<float>(), invM, total, C);
   }
   return {dx, sdyx, sdy};  // dweight = sum(dy*xhat), dbias = sum(dy)
 }
