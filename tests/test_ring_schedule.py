@@ -114,3 +114,17 @@ def test_segment_parts_partition():
                         prev_end[g] = off_r[g] + cnt_r[g]
                         covered[g] += cnt_r[g]
                 assert covered == cnt, (world, total, nrings, covered, cnt)
+
+
+def test_selector_mirrors_dispatch():
+    import torch
+    from torchmpi_amd.selector import select_allreduce
+    # small pow2 world -> recursive halving
+    s = select_allreduce(torch.zeros(1000), 4)
+    assert s.algorithm == "recursive-halving" and s.device == "cpu"
+    # large -> ring on CPU
+    s = select_allreduce(torch.zeros(1 << 22), 4)
+    assert s.algorithm == "ring"
+    # non-pow2 world: never RHD
+    s = select_allreduce(torch.zeros(100), 3)
+    assert s.algorithm == "ring"

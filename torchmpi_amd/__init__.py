@@ -398,6 +398,7 @@ class _AsyncNS:
 asynchronous = _AsyncNS()
 
 from . import nn  # noqa: E402
+from . import selector  # noqa: E402
 from . import ops  # noqa: E402
 from .parameterserver import ParameterServer  # noqa: E402
 
@@ -407,5 +408,5 @@ __all__ = [
     "allreduce", "reduce", "broadcast", "allgather", "reduce_scatter",
     "sendreceive", "send", "recv", "gather", "scatter", "alltoall",
     "asynchronous", "sync_handle", "Handle", "RedOp", "Communicator",
-    "hierarchical_allreduce", "ParameterServer", "nn", "ops",
+    "hierarchical_allreduce", "ParameterServer", "nn", "ops", "selector",
 ]
