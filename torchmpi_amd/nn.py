@@ -118,15 +118,29 @@ class GradientSynchronizer:
         for b in self.buckets:
             b["pending"] = len(b["params"])
             b["handle"] = None
+            b["ready"] = False
+        self._next_launch = 0
 
     # -- hot path ----------------------------------------------------------
+    # Buckets LAUNCH in fixed bucket-index order even if they become ready
+    # out of order: RCCL grouped-p2p ops match by per-communicator call
+    # order, so every rank must issue its bucket allreduces in the same
+    # sequence (the reference's per-communicator thread gave the same
+    # guarantee; DDP enforces the same rule).
 
     def _hook(self, p):
         bi = self._param_bucket[id(p)]
         b = self.buckets[bi]
         b["pending"] -= 1
         if b["pending"] == 0:
-            self._launch(b)
+            b["ready"] = True
+            self._launch_ready_in_order()
+
+    def _launch_ready_in_order(self):
+        while (self._next_launch < len(self.buckets)
+               and self.buckets[self._next_launch]["ready"]):
+            self._launch(self.buckets[self._next_launch])
+            self._next_launch += 1
 
     def _launch(self, b):
         from . import allreduce
@@ -135,14 +149,14 @@ class GradientSynchronizer:
         b["handle"] = allreduce(bucket.buffer, async_op=True, comm=self.comm)
 
     def finish(self):
-        """Flush un-launched buckets, wait for all allreduces (stream-level),
-        unpack with the 1/W average fused."""
+        """Flush un-launched buckets (fixed order), wait for all allreduces
+        (stream-level), unpack with the 1/W average fused."""
         if self.comm.size() == 1:
             return
         alpha = 1.0 / self.comm.size() if self.average else 1.0
         for b in self.buckets:
             if b["handle"] is None:
-                self._launch(b)  # params that never got a hook this step
+                self._launch(b)  # unused params / not-yet-ready buckets
         for b in self.buckets:
             b["handle"].wait()  # current stream waits on comm streams
             b["bucket"].unpack(alpha)
