@@ -133,6 +133,9 @@ PYBIND11_MODULE(_C, m) {
            [](CpuComm& c, at::Tensor input, at::Tensor output) {
              check_cpu_tensor(input);
              check_cpu_tensor(output);
+             TMA_CHECK(input.numel() % c.size() == 0 &&
+                           output.numel() == input.numel(),
+                       "alltoall: numel must be divisible by world_size");
              size_t bytes = static_cast<size_t>(input.numel()) * input.element_size() /
                             c.size();
              return c.enqueue([&c, input, output, bytes] {

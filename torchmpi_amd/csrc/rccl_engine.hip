@@ -657,6 +657,8 @@ class RcclEngine {
     check_gpu_tensor(output);
     auto work = std::make_shared<GpuWork>();
     work->keep = {input, output};
+    TMA_CHECK(input.numel() % world_ == 0 && output.numel() == input.numel(),
+              "alltoall: numel must be divisible by world_size and match");
     size_t bytes =
         static_cast<size_t>(input.numel()) * input.element_size() / world_;
     hipEvent_t ready = record_ready();

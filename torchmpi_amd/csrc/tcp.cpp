@@ -9,6 +9,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 
@@ -47,9 +48,18 @@ sockaddr_in parse_addr(const std::string& hostport) {
 }
 
 // Wait for fd to become ready for events (POLLIN/POLLOUT).
+int tcp_timeout_ms() {
+  static int t = [] {
+    const char* e = std::getenv("TORCHMPI_TCP_TIMEOUT_MS");
+    return e ? std::atoi(e) : 600000;  // generous: ranks can skew minutes
+                                       // during MIOpen warmup/find phases
+  }();
+  return t;
+}
+
 void wait_ready(int fd, short events) {
   pollfd p{fd, events, 0};
-  int r = poll(&p, 1, 120000);
+  int r = poll(&p, 1, tcp_timeout_ms());
   TMA_CHECK(r > 0, "tcp poll timeout/error (peer dead?)");
   TMA_CHECK(!(p.revents & (POLLERR | POLLHUP | POLLNVAL)) || (p.revents & events),
             "tcp socket error");
@@ -180,7 +190,7 @@ void TcpMesh::duplex(int speer, const void* sbuf, size_t sn, int rpeer, void* rb
       int np = 0;
       if (sent < sn) ps[np++] = {sfd, POLLOUT, 0};
       if (rcvd < rn) ps[np++] = {rfd, POLLIN, 0};
-      int r = poll(ps, np, 120000);
+      int r = poll(ps, np, tcp_timeout_ms());
       TMA_CHECK(r > 0, "tcp duplex poll timeout (deadlock or dead peer?)");
     }
   }
