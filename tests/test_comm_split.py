@@ -92,3 +92,23 @@ def _split_by_node_body(tm):
 
 def test_split_by_node():
     run_distributed(_split_by_node_body, world=2)
+
+
+def _hier_2x4_body(tm):
+    """BASELINE config #5 topology: two groups of four (intra), four pairs
+    (inter); hierarchical allreduce must equal the flat allreduce."""
+    W, r = tm.size(), tm.rank()
+    assert W == 8
+    intra = tm.split(color=r // 4, key=r)   # 2 x 4-rank groups
+    inter = tm.split(color=r % 4, key=r)    # 4 x 2-rank groups
+    assert intra.size() == 4 and inter.size() == 2
+    n = 4099  # not divisible by 4 -> unequal shards path
+    t = torch.arange(n, dtype=torch.float32) * (r + 1)
+    flat = t.clone()
+    tm.allreduce(flat)
+    tm.hierarchical_allreduce(t, intra, inter)
+    assert torch.allclose(t, flat), (t[:4], flat[:4])
+
+
+def test_hierarchical_2x4():
+    run_distributed(_hier_2x4_body, world=8)
