@@ -46,17 +46,30 @@ def verify_allreduce(tm, device):
     n = 1 << 20  # 4 MB fp32: multi-ring + graph eligible
     t = torch.empty(n, device=device)
     expect = float(W * (W + 1) / 2)
-    for it in range(3):
+    eng = tm.communicator().engine
+
+    def check():
         t.fill_(float(tm.rank() + 1))
         tm.allreduce(t)
         torch.cuda.synchronize()
-        ok = bool(torch.allclose(t, torch.full_like(t, expect)))
-        if not ok and it == 0:
-            raise RuntimeError("eager multi-ring allreduce produced wrong "
-                               "values — refusing to benchmark")
-        if not ok:
-            tm.communicator().engine.set_graphs_enabled(False)
-            return "graphs-disabled(replay-mismatch)"
+        return bool(torch.allclose(t, torch.full_like(t, expect)))
+
+    for it in range(3):
+        if check():
+            continue
+        if it == 0:
+            # eager multi-ring wrong: degrade to single ring + no graphs and
+            # re-verify — a slower bench beats no bench; still-wrong => raise.
+            eng.set_max_rings(1)
+            eng.set_graphs_enabled(False)
+            if all(check() for _ in range(2)):
+                return "degraded(single-ring,no-graphs)"
+            raise RuntimeError("allreduce produced wrong values even on the "
+                               "single-ring path — refusing to benchmark")
+        eng.set_graphs_enabled(False)
+        if not check():
+            raise RuntimeError("eager allreduce wrong after graph disable")
+        return "graphs-disabled(replay-mismatch)"
     return "ok"
 
 

@@ -198,6 +198,9 @@ class RcclEngine {
   int num_rings() const { return static_cast<int>(comms_.size()); }
   bool graphs_enabled() const { return graphs_enabled_; }
   void set_graphs_enabled(bool v) { graphs_enabled_ = v; }
+  // Cap the rings used per collective (bench-time fallback: a wrong
+  // multi-ring result degrades to single-ring instead of no result).
+  void set_max_rings(int v) { max_rings_ = v < 1 ? 1 : v; }
 
   // 1-GPU validation of the hipGraph path (tests/test_gpu_collectives.py):
   // captures {grouped self ncclSend/ncclRecv -> local reduce kernel} and
@@ -290,7 +293,7 @@ class RcclEngine {
     }
 
     const size_t align = 16 / es;  // keep 16B alignment for the reduce kernel
-    int nrings = count * es >= multiring_min_bytes_ ? num_rings() : 1;
+    int nrings = count * es >= multiring_min_bytes_ ? used_rings() : 1;
 
     std::vector<size_t> soff, scnt;
     split_aligned(count, nrings, align, soff, scnt);
@@ -363,7 +366,7 @@ class RcclEngine {
     hipEvent_t ready = record_ready();
     char* base = static_cast<char*>(input.data_ptr());
     const size_t align = 16 / es;
-    int nrings = count * es >= multiring_min_bytes_ ? num_rings() : 1;
+    int nrings = count * es >= multiring_min_bytes_ ? used_rings() : 1;
     // Ring r reduces part r of every global segment; the union of rank g's
     // parts is exactly its contiguous segment g.
     for (int r = 0; r < nrings; ++r) {
@@ -410,7 +413,7 @@ class RcclEngine {
     // Multi-chain: stripe the tensor over the link-disjoint rings; each
     // stripe flows down ring r''s chain (root first) chunk-pipelined.
     char* base = static_cast<char*>(t.data_ptr());
-    int nrings = bytes >= multiring_min_bytes_ ? num_rings() : 1;
+    int nrings = bytes >= multiring_min_bytes_ ? used_rings() : 1;
     std::vector<size_t> soff, scnt;
     split_aligned(bytes, nrings, 16, soff, scnt);
     for (int r = 0; r < nrings; ++r) {
@@ -538,7 +541,7 @@ class RcclEngine {
       cnt[g] = elems;
     }
     const size_t align = 16 / es2 ? 16 / es2 : 1;
-    int nrings = bytes >= multiring_min_bytes_ ? num_rings() : 1;
+    int nrings = bytes >= multiring_min_bytes_ ? used_rings() : 1;
     for (int r = 0; r < nrings; ++r) {
       std::vector<size_t> off_r, cnt_r;
       segment_parts(off, cnt, r, nrings, align, off_r, cnt_r);
@@ -1064,6 +1067,9 @@ class RcclEngine {
   std::vector<ncclComm_t> comms_;
   std::vector<c10::hip::HIPStreamMasqueradingAsCUDA> streams_;
   std::vector<int> pos_, next_, prev_;
+  int used_rings() const { return std::min(num_rings(), max_rings_); }
+
+  int max_rings_ = 1 << 30;
   hipEvent_t order_ev_ = nullptr;
   hipEvent_t join_ev_ = nullptr;
   size_t chunk_bytes_;
@@ -1197,6 +1203,7 @@ void register_gpu(pybind11::module_& m) {
       .def("num_rings", &RcclEngine::num_rings)
       .def("graphs_enabled", &RcclEngine::graphs_enabled)
       .def("set_graphs_enabled", &RcclEngine::set_graphs_enabled)
+      .def("set_max_rings", &RcclEngine::set_max_rings)
       .def("graph_selfcheck", &RcclEngine::graph_selfcheck,
            py::call_guard<py::gil_scoped_release>())
       .def("allreduce", &RcclEngine::allreduce, py::arg("tensor"), py::arg("op"),
