@@ -52,7 +52,11 @@ def verify_allreduce(tm, device):
         t.fill_(float(tm.rank() + 1))
         tm.allreduce(t)
         torch.cuda.synchronize()
-        return bool(torch.allclose(t, torch.full_like(t, expect)))
+        ok = bool(torch.allclose(t, torch.full_like(t, expect)))
+        # The verdict must be GLOBAL: every rank takes the same branch below
+        # (the fallback path issues extra collectives).
+        flags = tm.allgather(torch.tensor([1.0 if ok else 0.0]))
+        return bool(flags.min() > 0.5)
 
     for it in range(3):
         if check():
