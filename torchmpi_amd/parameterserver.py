@@ -27,10 +27,11 @@ _RULES = {"copy": 0, "add": 1}
 
 
 class PsHandle:
-    def __init__(self, work=None, event=None, post=None):
+    def __init__(self, work=None, event=None, post=None, keep=None):
         self._work = work
         self._event = event
         self._post = post
+        self._keep = keep  # tensors the in-flight op reads/writes
 
     def wait(self):
         if self._work is not None:
@@ -116,8 +117,8 @@ class ParameterServer:
             self._gpu.send(tid, src, _RULES[rule])
             ev = torch.cuda.Event()
             ev.record()
-            return PsHandle(event=ev)
-        return PsHandle(work=self._ctx.send(tid, src, _RULES[rule]))
+            return PsHandle(event=ev, keep=src)
+        return PsHandle(work=self._ctx.send(tid, src, _RULES[rule]), keep=src)
 
     def receive(self, tensor: torch.Tensor,
                 out: Optional[torch.Tensor] = None) -> PsHandle:
@@ -130,15 +131,15 @@ class ParameterServer:
             self._gpu.receive(tid, dst)
             ev = torch.cuda.Event()
             ev.record()
-            return PsHandle(event=ev)
+            return PsHandle(event=ev, keep=dst)
         flat = dst.contiguous()
         if flat.data_ptr() != dst.data_ptr():
             w = self._ctx.receive(tid, flat)
 
             def post():
                 dst.copy_(flat)
-            return PsHandle(work=w, post=post)
-        return PsHandle(work=self._ctx.receive(tid, flat))
+            return PsHandle(work=w, post=post, keep=(flat, dst))
+        return PsHandle(work=self._ctx.receive(tid, flat), keep=flat)
 
     prefetch = receive  # reference: prefetch == async receive
 
