@@ -145,6 +145,11 @@ class GradientSynchronizer:
     def _launch(self, b):
         from . import allreduce
         bucket = b["bucket"]
+        if b["params"][0].grad is not bucket.tensors[0]:
+            raise RuntimeError(
+                "GradientSynchronizer: grad tensors were reallocated — use "
+                "optimizer.zero_grad(set_to_none=False) so the registered "
+                "bucket buffers stay valid")
         bucket.pack()  # on current stream (autograd stream)
         b["handle"] = allreduce(bucket.buffer, async_op=True, comm=self.comm)
 
