@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "kernels", "reduce.hip"),
     os.path.join(CSRC, "kernels", "pack.hip"),
     os.path.join(CSRC, "kernels", "batchnorm.hip"),
+    os.path.join(CSRC, "kernels", "rmsnorm.hip"),
 ]
 
 extra_cxx = ["-O3", "-std=c++17", "-Wall"]

@@ -21,6 +21,7 @@ namespace tma {
 void register_gpu(py::module_& m);     // defined in rccl_engine.hip
 void register_ipc_ps(py::module_& m);  // defined in ipc_ps.hip
 void register_batchnorm(py::module_& m);  // defined in kernels/batchnorm.hip
+void register_rmsnorm(py::module_& m);    // defined in kernels/rmsnorm.hip
 
 namespace {
 
@@ -195,6 +196,7 @@ PYBIND11_MODULE(_C, m) {
   register_gpu(m);
   register_ipc_ps(m);
   register_batchnorm(m);
+  register_rmsnorm(m);
 }
 
 }  // namespace tma

@@ -42,6 +42,11 @@ class RMSNorm(nn.Module):
         self.weight = nn.Parameter(torch.ones(dim))
 
     def forward(self, x):
+        if x.is_cuda:
+            from ..ops.rmsnorm import fused_rmsnorm, fused_rmsnorm_supported
+            xc = x.contiguous()
+            if self.weight.dtype == torch.float32 and fused_rmsnorm_supported(xc):
+                return fused_rmsnorm(xc, self.weight, self.eps)
         dt = x.dtype
         x = x.float()
         x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
@@ -132,6 +137,10 @@ class LlamaModel(nn.Module):
                                      self.cfg.max_seq_len, self.cfg.rope_theta,
                                      tokens.device)
         x = self.embed(tokens)
+        if x.is_cuda and torch.is_autocast_enabled():
+            # bf16 residual stream (standard bf16 mixed precision; RMSNorm
+            # stats still accumulate in fp32 inside the fused kernel).
+            x = x.to(torch.bfloat16)
         for blk in self.blocks:
             x = blk(x, self.freqs)
         x = self.norm(x)
