@@ -160,6 +160,10 @@ class RcclEngine {
       std::memcpy(&id, uids[r].data(), sizeof(id));
       TMA_NCCL_CHECK(ncclCommInitRank(&comms_[r], world_, id, rank_));
       streams_.push_back(c10::hip::getStreamFromPoolMasqueradingAsCUDA(/*isHighPriority=*/true, device_));
+      recv_streams_.push_back(
+          c10::hip::getStreamFromPoolMasqueradingAsCUDA(true, device_));
+      red_streams_.push_back(
+          c10::hip::getStreamFromPoolMasqueradingAsCUDA(true, device_));
       // position of our rank in ring r, successor and predecessor
       const auto& perm = rings[r];
       TMA_CHECK(static_cast<int>(perm.size()) == world_, "ring size != world");
@@ -181,9 +185,17 @@ class RcclEngine {
     if (gm) graph_min_bytes_ = static_cast<size_t>(std::atoll(gm));
     const char* rh = std::getenv("TORCHMPI_RHD_MAX_BYTES");
     rhd_max_bytes_ = rh ? static_cast<size_t>(std::atoll(rh)) : (256u << 10);
+    const char* ov = std::getenv("TORCHMPI_RING_OVERLAP");
+    ring_overlap_ = ov && std::string(ov) == "1";
   }
 
   ~RcclEngine() {
+    for (auto ev : fork_ev_)
+      if (ev) (void)hipEventDestroy(ev);
+    for (auto& v : recv_done_)
+      for (auto ev : v) (void)hipEventDestroy(ev);
+    for (auto& v : red_done_)
+      for (auto ev : v) (void)hipEventDestroy(ev);
     if (order_ev_) (void)hipEventDestroy(order_ev_);
     if (join_ev_) (void)hipEventDestroy(join_ev_);
     for (auto& kv : graphs_) {
@@ -380,8 +392,12 @@ class RcclEngine {
       order_in(s);
       at::Tensor scratch = scratch_like(2 * maxc * es, s);
       work->keep.push_back(scratch);
-      ring_rs(r, base, off_r, cnt_r, d, op,
-              static_cast<char*>(scratch.data_ptr()));
+      if (ring_overlap_ && world_ > 2)
+        ring_rs_overlap(r, base, off_r, cnt_r, d, op,
+                        static_cast<char*>(scratch.data_ptr()));
+      else
+        ring_rs(r, base, off_r, cnt_r, d, op,
+                static_cast<char*>(scratch.data_ptr()));
       record_stream_uses(input, s);
       record_done(work, s);
     }
@@ -860,11 +876,91 @@ class RcclEngine {
     }
   }
 
+  // Overlapped ring reduce-scatter (TORCHMPI_RING_OVERLAP=1, experimental —
+  // flip on after measuring on the 8-GPU node): sends run on streams_[r],
+  // recvs on recv_streams_[r], local reductions on red_streams_[r], so the
+  // step-s reduction overlaps the step-s+1 transfer. Dependencies:
+  //   send(s)  waits reduce(s-1)   (it sends the segment reduced last step)
+  //   recv(s)  waits reduce(s-2)   (double-buffer slot reuse)
+  //   reduce(s) waits recv(s)
+  // Grouped-p2p matching holds: every rank issues send-group then recv-group
+  // per step, so per-peer per-direction call order is identical everywhere.
+  void ring_rs_overlap(int r, char* base, const std::vector<size_t>& off,
+                       const std::vector<size_t>& cnt, DType d, RedOp op,
+                       char* sc) {
+    const size_t es = dtype_size(d);
+    const int W = world_;
+    auto sA = streams_[r];
+    auto sB = recv_streams_[r];
+    auto sR = red_streams_[r];
+    const auto& perm = rings_[r];
+    size_t maxc = 0;
+    for (auto c : cnt) maxc = std::max(maxc, c);
+    const int p = pos_[r];
+    ensure_overlap_events(r, W - 1);
+    auto& er = recv_done_[r];
+    auto& dr = red_done_[r];
+    hipEvent_t e_in = fork_ev_[r];
+    TMA_HIP_CHECK(hipEventRecord(e_in, sA.stream()));
+    TMA_HIP_CHECK(hipStreamWaitEvent(sB.stream(), e_in, 0));
+    TMA_HIP_CHECK(hipStreamWaitEvent(sR.stream(), e_in, 0));
+    for (int step = 0; step < W - 1; ++step) {
+      int send_seg, recv_seg;
+      ring_step_segs(perm, W, p, step, /*phase=*/0, send_seg, recv_seg);
+      char* rbuf = sc + (step % 2) * maxc * es;
+      if (step > 0)
+        TMA_HIP_CHECK(hipStreamWaitEvent(sA.stream(), dr[step - 1], 0));
+      if (cnt[send_seg] > 0) {
+        TMA_NCCL_CHECK(ncclGroupStart());
+        TMA_NCCL_CHECK(ncclSend(base + off[send_seg] * es, cnt[send_seg] * es,
+                                ncclInt8, next_[r], comms_[r], sA.stream()));
+        TMA_NCCL_CHECK(ncclGroupEnd());
+      }
+      if (step >= 2)
+        TMA_HIP_CHECK(hipStreamWaitEvent(sB.stream(), dr[step - 2], 0));
+      if (cnt[recv_seg] > 0) {
+        TMA_NCCL_CHECK(ncclGroupStart());
+        TMA_NCCL_CHECK(ncclRecv(rbuf, cnt[recv_seg] * es, ncclInt8, prev_[r],
+                                comms_[r], sB.stream()));
+        TMA_NCCL_CHECK(ncclGroupEnd());
+      }
+      TMA_HIP_CHECK(hipEventRecord(er[step], sB.stream()));
+      TMA_HIP_CHECK(hipStreamWaitEvent(sR.stream(), er[step], 0));
+      if (cnt[recv_seg] > 0)
+        launch_reduce(base + off[recv_seg] * es, rbuf, cnt[recv_seg], d, op,
+                      sR.stream());
+      TMA_HIP_CHECK(hipEventRecord(dr[step], sR.stream()));
+    }
+    // Join the aux streams back into the ring's main stream (the allgather
+    // phase and order_out continue on streams_[r]).
+    TMA_HIP_CHECK(hipStreamWaitEvent(sA.stream(), dr[W - 2], 0));
+  }
+
+  void ensure_overlap_events(int r, int steps) {
+    if ((int)fork_ev_.size() <= r) {
+      fork_ev_.resize(num_rings(), nullptr);
+      recv_done_.resize(num_rings());
+      red_done_.resize(num_rings());
+    }
+    if (!fork_ev_[r])
+      TMA_HIP_CHECK(hipEventCreateWithFlags(&fork_ev_[r], hipEventDisableTiming));
+    while ((int)recv_done_[r].size() < steps) {
+      hipEvent_t a, b;
+      TMA_HIP_CHECK(hipEventCreateWithFlags(&a, hipEventDisableTiming));
+      TMA_HIP_CHECK(hipEventCreateWithFlags(&b, hipEventDisableTiming));
+      recv_done_[r].push_back(a);
+      red_done_[r].push_back(b);
+    }
+  }
+
   void ring_reduce_scatter_buf(int r, char* base, size_t count, DType d, RedOp op,
                                char* sc) {
     std::vector<size_t> off, cnt;
     split_segments(count, world_, off, cnt);
-    ring_rs(r, base, off, cnt, d, op, sc);
+    if (ring_overlap_ && world_ > 2)
+      ring_rs_overlap(r, base, off, cnt, d, op, sc);
+    else
+      ring_rs(r, base, off, cnt, d, op, sc);
   }
 
   // Recursive halving-doubling allreduce on comms_[0]/streams_[0]:
@@ -957,7 +1053,10 @@ class RcclEngine {
                           char* sc) {
     std::vector<size_t> off, cnt;
     split_segments(count, world_, off, cnt);
-    ring_rs(r, base, off, cnt, d, op, sc);
+    if (ring_overlap_ && world_ > 2)
+      ring_rs_overlap(r, base, off, cnt, d, op, sc);
+    else
+      ring_rs(r, base, off, cnt, d, op, sc);
     ring_ag(r, base, off, cnt, dtype_size(d));
   }
 
@@ -1070,6 +1169,10 @@ class RcclEngine {
   int used_rings() const { return std::min(num_rings(), max_rings_); }
 
   int max_rings_ = 1 << 30;
+  bool ring_overlap_ = false;
+  std::vector<hipEvent_t> fork_ev_;
+  std::vector<std::vector<hipEvent_t>> recv_done_, red_done_;
+  std::vector<c10::hip::HIPStreamMasqueradingAsCUDA> recv_streams_, red_streams_;
   hipEvent_t order_ev_ = nullptr;
   hipEvent_t join_ev_ = nullptr;
   size_t chunk_bytes_;
