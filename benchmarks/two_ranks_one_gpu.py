@@ -1,8 +1,15 @@
 #!/usr/bin/env python3
 """Experiment: run the full RCCL engine with world_size=2, both ranks on the
-SAME GPU (device 0). If RCCL accepts two ranks per device this validates the
-grouped send/recv ring schedule + local reduce end-to-end on a 1-GPU box.
-Exit code 0 + 'TWO_RANKS_OK' on success; prints the RCCL error otherwise.
+SAME GPU (device 0).
+
+FINDING (recorded 2026-09-13, ROCm 7.2 RCCL): ncclCommInitRank returns
+'invalid usage' for two ranks on one device — RCCL does not support
+multi-rank-per-GPU, so multi-rank engine paths cannot be exercised on a
+1-GPU box. Multi-rank coverage therefore comes from (a) the exported-
+schedule simulation tests (tests/test_ring_schedule.py), (b) the CPU
+transport running the same algorithms under real multi-process tests, and
+(c) bench.py's analytic self-check at startup on the real 8-GPU node.
+Kept for re-testing on future RCCL versions.
 """
 import os
 import sys

@@ -47,7 +47,6 @@ sockaddr_in parse_addr(const std::string& hostport) {
   return sa;
 }
 
-// Wait for fd to become ready for events (POLLIN/POLLOUT).
 int tcp_timeout_ms() {
   static int t = [] {
     const char* e = std::getenv("TORCHMPI_TCP_TIMEOUT_MS");
@@ -55,14 +54,6 @@ int tcp_timeout_ms() {
                                        // during MIOpen warmup/find phases
   }();
   return t;
-}
-
-void wait_ready(int fd, short events) {
-  pollfd p{fd, events, 0};
-  int r = poll(&p, 1, tcp_timeout_ms());
-  TMA_CHECK(r > 0, "tcp poll timeout/error (peer dead?)");
-  TMA_CHECK(!(p.revents & (POLLERR | POLLHUP | POLLNVAL)) || (p.revents & events),
-            "tcp socket error");
 }
 
 }  // namespace
@@ -194,7 +185,6 @@ void TcpMesh::duplex(int speer, const void* sbuf, size_t sn, int rpeer, void* rb
       TMA_CHECK(r > 0, "tcp duplex poll timeout (deadlock or dead peer?)");
     }
   }
-  (void)wait_ready;  // silence unused in some builds
 }
 
 }  // namespace tma
