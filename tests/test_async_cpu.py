@@ -63,3 +63,27 @@ def _ready_poll_body(tm):
 
 def test_ready_poll():
     run_distributed(_ready_poll_body, world=2)
+
+
+def _async_namespace_body(tm):
+    # reference mpi.async.* namespace parity + sync_handle
+    W, r = tm.size(), tm.rank()
+    t = torch.full((50,), float(r + 1))
+    h = tm.asynchronous.allreduce(t)
+    tm.sync_handle(h)
+    assert torch.equal(t, torch.full((50,), float(W * (W + 1) / 2)))
+    b = torch.arange(5.0) if r == 0 else torch.zeros(5)
+    tm.sync_handle(tm.asynchronous.broadcast(b, root=0))
+    assert torch.equal(b, torch.arange(5.0))
+    h, out = tm.asynchronous.allgather(torch.full((3,), float(r)))
+    h.wait()
+    for k in range(W):
+        assert torch.equal(out[k], torch.full((3,), float(k)))
+    s = torch.full((4,), float(r))
+    rv = torch.zeros(4)
+    tm.sync_handle(tm.asynchronous.sendreceive(s, (r + 1) % W, rv, (r - 1) % W))
+    assert torch.equal(rv, torch.full((4,), float((r - 1) % W)))
+
+
+def test_async_namespace():
+    run_distributed(_async_namespace_body, world=2)
