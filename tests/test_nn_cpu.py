@@ -129,3 +129,39 @@ def _e2e_training_body(tm):
 
 def test_e2e_dp_training_matches_single_process():
     run_distributed(_e2e_training_body, world=2)
+
+
+def _bf16_comm_dtype_body(tm):
+    """The bench's actual configuration: fp32 params, bf16 bucket compression
+    (cast-fused pack/unpack), averaged — CPU-path equivalence with tolerance."""
+    from torchmpi_amd.nn import GradientSynchronizer
+    W, r = tm.size(), tm.rank()
+    model = _make_model(seed=3)
+    sync = GradientSynchronizer(model, average=True, comm_dtype=torch.bfloat16)
+    torch.manual_seed(50 + r)
+    x = torch.randn(8, 16)
+    y = torch.randn(8, 4)
+    loss = torch.nn.functional.mse_loss(model(x), y)
+    loss.backward()
+    fp32_grads = [p.grad.clone() for p in model.parameters()]
+    # reference: average of each rank's fp32 grads (recompute locally)
+    ref_model = _make_model(seed=3)
+    acc = None
+    for k in range(W):
+        torch.manual_seed(50 + k)
+        xk, yk = torch.randn(8, 16), torch.randn(8, 4)
+        ref_model.zero_grad()
+        torch.nn.functional.mse_loss(ref_model(xk), yk).backward()
+        g = [p.grad.clone() for p in ref_model.parameters()]
+        acc = g if acc is None else [a + b for a, b in zip(acc, g)]
+    expect = [a / W for a in acc]
+    sync.finish()
+    for p, e in zip(model.parameters(), expect):
+        # bf16 wire: ~3 rounding steps (pack, reduce, unpack)
+        assert torch.allclose(p.grad, e, rtol=0.1, atol=0.05), \
+            (p.grad.flatten()[:3], e.flatten()[:3])
+
+
+
+def test_bf16_comm_dtype():
+    run_distributed(_bf16_comm_dtype_body, world=2)
