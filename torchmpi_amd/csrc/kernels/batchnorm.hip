@@ -76,6 +76,10 @@ VecGeom vec_geom(int C, int VEC, size_t M) {
   int rows_per_iter = 256 / g.c8_per_block;
   int max_rb = (int)((M + rows_per_iter - 1) / rows_per_iter);
   g.row_blocks = std::min(std::max(4096 / g.c_blocks, 1), std::max(max_rb, 1));
+  // Small-M/large-C shapes: keep the partial buffers (row_blocks * C * 8 B)
+  // under half the data read, or the reduction traffic dominates.
+  int traffic_cap = std::max(1, (int)(M / VEC));
+  g.row_blocks = std::min(g.row_blocks, traffic_cap);
   g.grid = g.c_blocks * g.row_blocks;
   g.ok = true;
   return g;
