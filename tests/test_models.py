@@ -41,3 +41,19 @@ def test_bucket_cpu_roundtrip():
     b.unpack(alpha=0.5)
     for t, o in zip(ts, orig):
         assert torch.allclose(t, 0.5 * o)
+
+
+def test_llama3_8b_param_count():
+    """The benchmark model must be the real Llama-3-8B shape: 8.03B params
+    (computed in closed form from the config; the 8B model is only ever
+    instantiated on the 288 GB GPU)."""
+    from torchmpi_amd.models import llama3_8b
+    c = llama3_8b()
+    hd = c.dim // c.n_heads
+    attn = (c.dim * c.n_heads * hd + 2 * c.dim * c.n_kv_heads * hd
+            + c.n_heads * hd * c.dim)
+    mlp = 2 * c.dim * c.ffn_dim + c.ffn_dim * c.dim
+    per_layer = attn + mlp + 2 * c.dim
+    total = (c.vocab_size * c.dim + c.n_layers * per_layer + c.dim
+             + c.dim * c.vocab_size)
+    assert abs(total - 8.03e9) < 0.01e9, total
