@@ -251,7 +251,7 @@ void PsContext::loop() {
     // same sockets; safe because this thread is the only reader).
     for (size_t i = 1; i < fds.size(); ++i) fds[i].fd = mesh_fd(peer_of[i]);
     int rc = poll(fds.data(), fds.size(), 500);
-    if (rc <= 0) continue;
+    if (rc <= 0) continue;  // timeout tick or EINTR: re-check running_
     if (fds[0].revents & POLLIN) {
       char b;
       (void)!read(wake_pipe_[0], &b, 1);

@@ -113,7 +113,10 @@ void TcpMesh::connect_mesh(int rank, const std::vector<std::string>& addrs) {
 
   // Accept one connection from every higher rank.
   for (int i = rank + 1; i < world; ++i) {
-    int fd = accept(listen_fd_, nullptr, nullptr);
+    int fd;
+    do {
+      fd = accept(listen_fd_, nullptr, nullptr);
+    } while (fd < 0 && errno == EINTR);
     TMA_CHECK(fd >= 0, "accept failed");
     set_sockopts(fd);
     uint32_t peer = 0;
@@ -158,7 +161,8 @@ void TcpMesh::duplex(int speer, const void* sbuf, size_t sn, int rpeer, void* rb
       if (r > 0) {
         sent += static_cast<size_t>(r);
         progressed = true;
-      } else if (r < 0 && errno != EAGAIN && errno != EWOULDBLOCK) {
+      } else if (r < 0 && errno != EAGAIN && errno != EWOULDBLOCK &&
+                 errno != EINTR) {
         throw std::runtime_error("torchmpi_amd: tcp send failed: " +
                                  std::string(strerror(errno)));
       }
@@ -170,7 +174,7 @@ void TcpMesh::duplex(int speer, const void* sbuf, size_t sn, int rpeer, void* rb
         progressed = true;
       } else if (r == 0) {
         throw std::runtime_error("torchmpi_amd: peer closed connection");
-      } else if (errno != EAGAIN && errno != EWOULDBLOCK) {
+      } else if (errno != EAGAIN && errno != EWOULDBLOCK && errno != EINTR) {
         throw std::runtime_error("torchmpi_amd: tcp recv failed: " +
                                  std::string(strerror(errno)));
       }
@@ -182,6 +186,7 @@ void TcpMesh::duplex(int speer, const void* sbuf, size_t sn, int rpeer, void* rb
       if (sent < sn) ps[np++] = {sfd, POLLOUT, 0};
       if (rcvd < rn) ps[np++] = {rfd, POLLIN, 0};
       int r = poll(ps, np, tcp_timeout_ms());
+      if (r < 0 && errno == EINTR) continue;  // signal during poll: retry
       TMA_CHECK(r > 0, "tcp duplex poll timeout (deadlock or dead peer?)");
     }
   }
