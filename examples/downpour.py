@@ -37,7 +37,8 @@ def main_worker(pull_every=4):
 
     torch.manual_seed(4321 + rank)
     pending = None
-    for step in range(40):
+    n_steps = 10 if os.environ.get("TORCHMPI_EXAMPLE_FAST") == "1" else 40
+    for step in range(n_steps):
         x = torch.randn(32, 1, 8, 8).view(32, -1)
         y = torch.randint(0, 10, (32,))
         model.zero_grad()

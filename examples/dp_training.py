@@ -53,12 +53,16 @@ def main_worker(checkpoint=None, resume=False):
     sync = GradientSynchronizer(model, average=True)
 
     # Synthetic "dataset": 4096 samples, rank-sharded sample_index % W.
+    # TORCHMPI_EXAMPLE_FAST=1 shrinks the run (CI under load).
+    fast = os.environ.get("TORCHMPI_EXAMPLE_FAST") == "1"
+    n_samples = 1024 if fast else 4096
+    n_epochs = 1 if fast else 3
     torch.manual_seed(999)
-    data = torch.randn(4096, 1, 28, 28)
-    labels = (data.view(4096, -1).sum(-1) * 10).long().abs() % 10
-    my_idx = torch.arange(rank, 4096, W)
+    data = torch.randn(n_samples, 1, 28, 28)
+    labels = (data.view(n_samples, -1).sum(-1) * 10).long().abs() % 10
+    my_idx = torch.arange(rank, n_samples, W)
 
-    for epoch in range(start_epoch, 3):
+    for epoch in range(start_epoch, n_epochs):
         perm = torch.randperm(len(my_idx))
         total_loss, n = 0.0, 0
         for i in range(0, len(my_idx), 64):

@@ -34,7 +34,8 @@ def main_worker(alpha=0.5, tau=4):
 
     opt = torch.optim.SGD(model.parameters(), lr=0.05, momentum=0.9)
     torch.manual_seed(100 + rank)
-    for step in range(40):
+    n_steps = 12 if os.environ.get("TORCHMPI_EXAMPLE_FAST") == "1" else 40
+    for step in range(n_steps):
         x = torch.randn(64, 32)
         y = torch.randint(0, 4, (64,))
         opt.zero_grad()
