@@ -61,7 +61,7 @@ def create_rendezvous(rank=None, world_size=None, master_addr=None,
         master_port,
         world_size,
         is_master=(rank == 0),
-        timeout=datetime.timedelta(seconds=300),
+        timeout=datetime.timedelta(seconds=600),
         wait_for_workers=False,
         use_libuv=True,
     )
