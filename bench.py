@@ -184,7 +184,10 @@ def main():
             "local_batch": local_batch,
             "image_size": img,
             "parallelism": f"dp{W}",
-            "overlap": "bucketed async allreduce over backward",
+            # At W=1 no collective runs (GradientSynchronizer.finish()
+            # early-returns): the number is a model+kernel throughput anchor.
+            "overlap": ("bucketed async allreduce over backward" if W > 1
+                        else "none (single GPU: no collectives in step)"),
         }
         if busbw is not None:
             cfg["allreduce_busbw_256MB_bf16_GBps"] = busbw

@@ -98,10 +98,17 @@ def _reduce_scatter_body(tm):
     cnt = base + (1 if r < rem else 0)
     off = r * base + min(r, rem)
     out = torch.zeros(cnt)
+    t_orig = t.clone()
     tm.reduce_scatter(t, out)
     s = W * (W + 1) / 2
     expect = torch.arange(off, off + cnt, dtype=torch.float32) * s
     assert torch.allclose(out, expect), (out[:4], expect[:4])
+    # MPI semantics: the input is preserved (round-1 verdict item 6).
+    assert torch.equal(t, t_orig), "reduce_scatter clobbered its input"
+    # In-place-workspace variant gives the same result, clobbering t.
+    out2 = torch.zeros(cnt)
+    tm.reduce_scatter_(t, out2)
+    assert torch.allclose(out2, expect)
 
 
 def test_reduce_scatter():

@@ -70,6 +70,22 @@ def test_hierarchical_allreduce():
     run_distributed(_hierarchical_body, world=4)
 
 
+def _hierarchical_avg_body(tm):
+    W, r = tm.size(), tm.rank()
+    intra = tm.split(color=r // 2, key=r)
+    inter = tm.split(color=r % 2, key=r)
+    t = torch.arange(1000, dtype=torch.float32) + (r + 1)
+    tm.hierarchical_allreduce(t, intra, inter, average=True)
+    # mean over ALL wi*we ranks (regression: was scaled only by 1/inter.size)
+    expect = torch.arange(1000, dtype=torch.float32) + \
+        sum(k + 1 for k in range(W)) / W
+    assert torch.allclose(t, expect), (t[:4], expect[:4])
+
+
+def test_hierarchical_allreduce_average():
+    run_distributed(_hierarchical_avg_body, world=4)
+
+
 def _nested_split_body(tm):
     W, r = tm.size(), tm.rank()
     half = tm.split(color=r // 2, key=r)

@@ -92,6 +92,61 @@ def test_gradient_synchronizer_overlap():
     run_distributed(_overlap_engine_body, world=2)
 
 
+def _accumulation_body(tm):
+    """no_sync() gradient accumulation: K-1 local backwards + 1 synced
+    backward == allreduce of the K-batch summed grads."""
+    from torchmpi_amd.nn import GradientSynchronizer
+    W, r = tm.size(), tm.rank()
+    K = 3
+    model = _make_model(seed=0)
+    sync = GradientSynchronizer(model, average=False)
+
+    # reference: sum grads of all K*W microbatches on one replica
+    ref_model = _make_model(seed=0)
+    for k in range(W * K):
+        torch.manual_seed(500 + k)
+        xk, yk = torch.randn(8, 16), torch.randn(8, 4)
+        torch.nn.functional.mse_loss(ref_model(xk), yk).backward()
+
+    for k in range(K):
+        torch.manual_seed(500 + r * K + k)
+        x, y = torch.randn(8, 16), torch.randn(8, 4)
+        loss = torch.nn.functional.mse_loss(model(x), y)
+        if k < K - 1:
+            with sync.no_sync():
+                loss.backward()
+        else:
+            loss.backward()
+    sync.finish()
+    for p, q in zip(model.parameters(), ref_model.parameters()):
+        assert torch.allclose(p.grad, q.grad, rtol=1e-4, atol=1e-6), \
+            (p.grad.flatten()[:3], q.grad.flatten()[:3])
+
+
+def test_gradient_accumulation_no_sync():
+    run_distributed(_accumulation_body, world=2)
+
+
+def _double_backward_error_body(tm):
+    """Accumulating without no_sync() must raise loudly, not silently
+    compute wrong gradients (advisor finding, round 1)."""
+    from torchmpi_amd.nn import GradientSynchronizer
+    model = _make_model(seed=0)
+    GradientSynchronizer(model, average=False)
+    x, y = torch.randn(8, 16), torch.randn(8, 4)
+    torch.nn.functional.mse_loss(model(x), y).backward()
+    try:
+        torch.nn.functional.mse_loss(model(x), y).backward()
+    except RuntimeError as e:
+        assert "no_sync" in str(e), e
+        return
+    raise AssertionError("second backward without no_sync did not raise")
+
+
+def test_double_backward_raises():
+    run_distributed(_double_backward_error_body, world=2)
+
+
 def _e2e_training_body(tm):
     """End-to-end DP-SGD equivalence (reference stack 3.5, examples/mnist):
     W-rank data-parallel training with averaged grads == single-process

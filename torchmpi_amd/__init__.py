@@ -286,16 +286,28 @@ def allgather(tensor: torch.Tensor, out: Optional[torch.Tensor] = None,
 
 
 @_traced
-def reduce_scatter(tensor: torch.Tensor, out: torch.Tensor, op="sum",
-                   async_op: bool = False, comm: Optional[Communicator] = None):
-    """Reduce `tensor` across ranks and scatter segment `rank` into `out`.
-    NOTE: clobbers `tensor` (it is the reduction workspace)."""
+def reduce_scatter_(tensor: torch.Tensor, out: torch.Tensor, op="sum",
+                    async_op: bool = False, comm: Optional[Communicator] = None):
+    """In-place-workspace variant: reduces `tensor` across ranks and scatters
+    segment `rank` into `out`, CLOBBERING `tensor` (it is the reduction
+    workspace). Zero-copy — use when the input is disposable (e.g. the
+    hierarchical allreduce's flat buffer)."""
     c = comm or communicator()
     if tensor.is_cuda:
         return _finish(Handle(c.engine.reduce_scatter(tensor, out, _op(op)), True),
                        async_op)
     return _finish(Handle(c.cpu.reduce_scatter(tensor, out, _op(op)), False),
                    async_op)
+
+
+@_traced
+def reduce_scatter(tensor: torch.Tensor, out: torch.Tensor, op="sum",
+                   async_op: bool = False, comm: Optional[Communicator] = None):
+    """Reduce `tensor` across ranks and scatter segment `rank` into `out`.
+    Input-preserving (MPI_Reduce_scatter semantics): the reduction runs on a
+    private copy, one extra D2D/memcpy of the input."""
+    work = tensor.clone()
+    return reduce_scatter_(work, out, op, async_op=async_op, comm=comm)
 
 
 @_traced
@@ -406,6 +418,7 @@ __all__ = [
     "start", "stop", "is_started", "rank", "size", "barrier", "world",
     "communicator", "split", "push_communicator", "pop_communicator",
     "allreduce", "reduce", "broadcast", "allgather", "reduce_scatter",
+    "reduce_scatter_",
     "sendreceive", "send", "recv", "gather", "scatter", "alltoall",
     "asynchronous", "sync_handle", "Handle", "RedOp", "Communicator",
     "hierarchical_allreduce", "ParameterServer", "nn", "ops", "selector",

@@ -55,6 +55,29 @@ __global__ __launch_bounds__(kBlock) void copy_bytes_kernel(
   if (tail < n) dst[tail] = src[tail];
 }
 
+// Fallback when src/dst are not mutually 16-byte aligned (e.g. a shard
+// offset into the client's full tensor that is not a multiple of 16).
+__global__ __launch_bounds__(kBlock) void copy_bytes_scalar_kernel(
+    char* __restrict__ dst, const char* __restrict__ src, size_t n) {
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    dst[i] = src[i];
+  }
+}
+
+inline void launch_copy_bytes(char* dst, const char* src, size_t n, int grid,
+                              hipStream_t stream) {
+  bool aligned = ((reinterpret_cast<uintptr_t>(dst) |
+                   reinterpret_cast<uintptr_t>(src)) & 15) == 0;
+  if (aligned)
+    hipLaunchKernelGGL(copy_bytes_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                       dst, src, n);
+  else
+    hipLaunchKernelGGL(copy_bytes_scalar_kernel, dim3(grid), dim3(kBlock), 0,
+                       stream, dst, src, n);
+}
+
 // System-scope atomic add: correct for concurrent updates from several GPUs
 // over the coherent xGMI fabric (shards are fine-grained allocations).
 __global__ __launch_bounds__(kBlock) void atomic_add_f32_kernel(
@@ -74,6 +97,72 @@ __global__ __launch_bounds__(kBlock) void atomic_add_f64_kernel(
        i += stride) {
     __hip_atomic_fetch_add(dst + i, src[i], __ATOMIC_RELAXED,
                            __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+}
+
+// 16-bit float 'add' via 32-bit system-scope CAS (no native 16-bit global
+// atomics across xGMI): each thread owns one aligned word = two elements.
+// The shard base is allocation-aligned and its byte size is rounded up to
+// 16, so the odd-count tail word is always in bounds.
+template <int DT>  // 0 = bf16, 1 = f16
+__device__ __forceinline__ float h16_to_f(uint16_t v) {
+  if (DT == 0) {
+    union { uint32_t u; float f; } c;
+    c.u = (uint32_t)v << 16;
+    return c.f;
+  }
+  __half_raw r;
+  r.x = v;
+  return __half2float(__half(r));
+}
+
+template <int DT>
+__device__ __forceinline__ uint16_t f_to_h16(float f) {
+  if (DT == 0) {
+    union { uint32_t u; float f; } c;
+    c.f = f;
+    uint32_t rnd = 0x7fff + ((c.u >> 16) & 1);  // round-to-nearest-even
+    return (uint16_t)((c.u + rnd) >> 16);
+  }
+  __half_raw r = __half_raw(__float2half(f));
+  return r.x;
+}
+
+template <int DT>
+__global__ __launch_bounds__(kBlock) void atomic_add_16_kernel(
+    uint32_t* __restrict__ dstw, const uint16_t* __restrict__ src, size_t n) {
+  const size_t nwords = n / 2;
+  const size_t stride = (size_t)gridDim.x * blockDim.x;
+  const size_t tid = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (size_t w = tid; w < nwords; w += stride) {
+    float a0 = h16_to_f<DT>(src[2 * w]);
+    float a1 = h16_to_f<DT>(src[2 * w + 1]);
+    uint32_t old = __hip_atomic_load(dstw + w, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_SYSTEM);
+    uint32_t desired;
+    do {
+      uint16_t lo = (uint16_t)(old & 0xffffu);
+      uint16_t hi = (uint16_t)(old >> 16);
+      desired = (uint32_t)f_to_h16<DT>(h16_to_f<DT>(lo) + a0) |
+                ((uint32_t)f_to_h16<DT>(h16_to_f<DT>(hi) + a1) << 16);
+    } while (!__hip_atomic_compare_exchange_strong(
+        dstw + w, &old, desired, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+        __HIP_MEMORY_SCOPE_SYSTEM));
+  }
+  if ((n & 1) && tid == 0) {
+    // last element occupies the low half of word `nwords`; preserve the high
+    // half (allocation padding, never read).
+    float a = h16_to_f<DT>(src[n - 1]);
+    uint32_t old = __hip_atomic_load(dstw + nwords, __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_SYSTEM);
+    uint32_t desired;
+    do {
+      uint16_t lo = (uint16_t)(old & 0xffffu);
+      desired = (old & 0xffff0000u) |
+                (uint32_t)f_to_h16<DT>(h16_to_f<DT>(lo) + a);
+    } while (!__hip_atomic_compare_exchange_strong(
+        dstw + nwords, &old, desired, __ATOMIC_RELAXED, __ATOMIC_RELAXED,
+        __HIP_MEMORY_SCOPE_SYSTEM));
   }
 }
 
@@ -114,7 +203,9 @@ class IpcPs {
     r.dtype = at_dtype(full_init);
     split_segments(r.count, world_, r.off, r.cnt);
     size_t es = dtype_size(r.dtype);
-    size_t bytes = std::max<size_t>(r.cnt[rank_] * es, 16);
+    // Round up to 16 B: keeps the 16-bit CAS 'add' tail word and the
+    // vectorized copy in bounds for any element count.
+    size_t bytes = std::max<size_t>((r.cnt[rank_] * es + 15) & ~size_t(15), 16);
     TMA_HIP_CHECK(hipExtMallocWithFlags(&r.own_shard, bytes,
                                         hipDeviceMallocFinegrained));
     TMA_HIP_CHECK(hipMemcpy(r.own_shard,
@@ -158,9 +249,8 @@ class IpcPs {
       char* src = base + r.off[p] * es;
       void* dst = r.peer_shards.at(p);
       if (rule == 0) {
-        hipLaunchKernelGGL(copy_bytes_kernel, dim3(ps_grid(r.cnt[p] * es / 16)),
-                           dim3(kBlock), 0, stream, static_cast<char*>(dst), src,
-                           r.cnt[p] * es);
+        launch_copy_bytes(static_cast<char*>(dst), src, r.cnt[p] * es,
+                          ps_grid(r.cnt[p] * es / 16), stream);
       } else if (r.dtype == DType::F32) {
         hipLaunchKernelGGL(atomic_add_f32_kernel, dim3(ps_grid(r.cnt[p])),
                            dim3(kBlock), 0, stream, static_cast<float*>(dst),
@@ -169,8 +259,16 @@ class IpcPs {
         hipLaunchKernelGGL(atomic_add_f64_kernel, dim3(ps_grid(r.cnt[p])),
                            dim3(kBlock), 0, stream, static_cast<double*>(dst),
                            reinterpret_cast<double*>(src), r.cnt[p]);
+      } else if (r.dtype == DType::BF16) {
+        hipLaunchKernelGGL(atomic_add_16_kernel<0>, dim3(ps_grid(r.cnt[p] / 2 + 1)),
+                           dim3(kBlock), 0, stream, static_cast<uint32_t*>(dst),
+                           reinterpret_cast<uint16_t*>(src), r.cnt[p]);
+      } else if (r.dtype == DType::F16) {
+        hipLaunchKernelGGL(atomic_add_16_kernel<1>, dim3(ps_grid(r.cnt[p] / 2 + 1)),
+                           dim3(kBlock), 0, stream, static_cast<uint32_t*>(dst),
+                           reinterpret_cast<uint16_t*>(src), r.cnt[p]);
       } else {
-        TMA_CHECK(false, "ps GPU 'add' rule supports f32/f64 shards");
+        TMA_CHECK(false, "ps GPU 'add' rule supports f32/f64/bf16/f16 shards");
       }
     }
   }
@@ -184,10 +282,9 @@ class IpcPs {
     char* base = static_cast<char*>(data.data_ptr());
     for (int p = 0; p < world_; ++p) {
       if (r.cnt[p] == 0) continue;
-      hipLaunchKernelGGL(copy_bytes_kernel, dim3(ps_grid(r.cnt[p] * es / 16)),
-                         dim3(kBlock), 0, stream, base + r.off[p] * es,
-                         static_cast<const char*>(r.peer_shards.at(p)),
-                         r.cnt[p] * es);
+      launch_copy_bytes(base + r.off[p] * es,
+                        static_cast<const char*>(r.peer_shards.at(p)),
+                        r.cnt[p] * es, ps_grid(r.cnt[p] * es / 16), stream);
     }
   }
 

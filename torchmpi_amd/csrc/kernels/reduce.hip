@@ -135,6 +135,33 @@ __global__ __launch_bounds__(kBlock) void scale_vec_kernel(T* __restrict__ data,
         from_acc<T>(to_acc(data[tail]) * static_cast<typename AccOf<T>::type>(alpha));
 }
 
+// f64 keeps alpha in double; other types take float (AccOf is float/f32).
+__global__ __launch_bounds__(kBlock) void scale_vec_f64_kernel(double* __restrict__ data,
+                                                               size_t n, double alpha) {
+  const size_t nvec = n / 2;
+  const size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  size_t i = static_cast<size_t>(blockIdx.x) * blockDim.x + threadIdx.x;
+  for (size_t v = i; v < nvec; v += stride) {
+    double2 dv = reinterpret_cast<double2*>(data)[v];
+    dv.x *= alpha;
+    dv.y *= alpha;
+    reinterpret_cast<double2*>(data)[v] = dv;
+  }
+  size_t tail = nvec * 2 + i;
+  if (tail < n) data[tail] *= alpha;
+}
+
+// Unaligned fallback: scalar grid-stride (mirrors launch_reduce's dispatch).
+template <typename T, typename A>
+__global__ __launch_bounds__(kBlock) void scale_scalar_kernel(T* __restrict__ data,
+                                                              size_t n, A alpha) {
+  const size_t stride = static_cast<size_t>(gridDim.x) * blockDim.x;
+  for (size_t i = static_cast<size_t>(blockIdx.x) * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    data[i] = from_acc<T>(to_acc(data[i]) * static_cast<typename AccOf<T>::type>(alpha));
+  }
+}
+
 // Measured on MI355X (benchmarks/probe_reduce.hip, profiles/): at >= 64 MB
 // streamed, 512 blocks (2/CU) beats 2048 by ~8% (less DVFS/L2 pressure);
 // below that the larger grid wins slightly.
@@ -191,23 +218,48 @@ void launch_reduce(void* dst, const void* src, size_t count, DType d, RedOp op,
 
 void launch_scale(void* data, size_t count, DType d, double alpha, hipStream_t stream) {
   int grid = grid_for(count / 4 + 1, 2 * count * dtype_size(d));
+  // A storage-offset view can be contiguous but not 16-byte aligned; the
+  // vector kernels require 16 B (same dispatch rule as launch_reduce).
+  bool aligned = (reinterpret_cast<uintptr_t>(data) & 15) == 0;
   switch (d) {
     case DType::F32:
-      hipLaunchKernelGGL((scale_vec_kernel<float>), dim3(grid), dim3(kBlock), 0, stream,
-                         static_cast<float*>(data), count, static_cast<float>(alpha));
+      if (aligned)
+        hipLaunchKernelGGL((scale_vec_kernel<float>), dim3(grid), dim3(kBlock), 0,
+                           stream, static_cast<float*>(data), count,
+                           static_cast<float>(alpha));
+      else
+        hipLaunchKernelGGL((scale_scalar_kernel<float, float>), dim3(grid),
+                           dim3(kBlock), 0, stream, static_cast<float*>(data), count,
+                           static_cast<float>(alpha));
       break;
     case DType::F64:
-      hipLaunchKernelGGL((scale_vec_kernel<double>), dim3(grid), dim3(kBlock), 0, stream,
-                         static_cast<double*>(data), count, static_cast<float>(alpha));
+      if (aligned)
+        hipLaunchKernelGGL(scale_vec_f64_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                           static_cast<double*>(data), count, alpha);
+      else
+        hipLaunchKernelGGL((scale_scalar_kernel<double, double>), dim3(grid),
+                           dim3(kBlock), 0, stream, static_cast<double*>(data), count,
+                           alpha);
       break;
     case DType::BF16:
-      hipLaunchKernelGGL((scale_vec_kernel<__hip_bfloat16>), dim3(grid), dim3(kBlock), 0,
-                         stream, static_cast<__hip_bfloat16*>(data), count,
-                         static_cast<float>(alpha));
+      if (aligned)
+        hipLaunchKernelGGL((scale_vec_kernel<__hip_bfloat16>), dim3(grid), dim3(kBlock),
+                           0, stream, static_cast<__hip_bfloat16*>(data), count,
+                           static_cast<float>(alpha));
+      else
+        hipLaunchKernelGGL((scale_scalar_kernel<__hip_bfloat16, float>), dim3(grid),
+                           dim3(kBlock), 0, stream, static_cast<__hip_bfloat16*>(data),
+                           count, static_cast<float>(alpha));
       break;
     case DType::F16:
-      hipLaunchKernelGGL((scale_vec_kernel<__half>), dim3(grid), dim3(kBlock), 0, stream,
-                         static_cast<__half*>(data), count, static_cast<float>(alpha));
+      if (aligned)
+        hipLaunchKernelGGL((scale_vec_kernel<__half>), dim3(grid), dim3(kBlock), 0,
+                           stream, static_cast<__half*>(data), count,
+                           static_cast<float>(alpha));
+      else
+        hipLaunchKernelGGL((scale_scalar_kernel<__half, float>), dim3(grid),
+                           dim3(kBlock), 0, stream, static_cast<__half*>(data), count,
+                           static_cast<float>(alpha));
       break;
     default:
       throw std::runtime_error("torchmpi_amd: scale supports float dtypes only");

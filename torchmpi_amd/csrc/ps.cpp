@@ -1,6 +1,8 @@
 #include "ps.h"
 
+#include <errno.h>
 #include <poll.h>
+#include <sys/socket.h>
 #include <unistd.h>
 
 #include <cstring>
@@ -33,8 +35,10 @@ void PsContext::start(int rank, const std::vector<std::string>& addrs) {
   rank_ = rank;
   world_ = static_cast<int>(addrs.size());
   if (world_ > 1) mesh_.connect_mesh(rank, addrs);
-  sock_m_.clear();
-  for (int i = 0; i < world_; ++i) sock_m_.push_back(std::make_unique<std::mutex>());
+  outq_.clear();
+  outq_.resize(world_);
+  out_bytes_ = 0;
+  dead_ = false;
   TMA_CHECK(pipe(wake_pipe_) == 0, "pipe failed");
   running_ = true;
   if (world_ > 1) loop_thread_ = std::thread([this] { loop(); });
@@ -52,6 +56,79 @@ void PsContext::stop() {
       close(wake_pipe_[i]);
       wake_pipe_[i] = -1;
     }
+  // Anything still pending can never complete once the loop is gone.
+  fail_all("parameter server stopped");
+}
+
+void PsContext::fail_all(const std::string& why) {
+  dead_ = true;
+  std::unordered_map<uint64_t, std::shared_ptr<Pending>> doomed;
+  {
+    std::lock_guard<std::mutex> lk(pend_m_);
+    doomed.swap(pending_);
+  }
+  for (auto& kv : doomed) {
+    kv.second->work->finish(std::make_exception_ptr(
+        std::runtime_error("torchmpi_amd ps: " + why)));
+  }
+  out_cv_.notify_all();  // release backpressure waiters (they re-check dead_)
+}
+
+void PsContext::enqueue_frame(int peer, const void* hdr, size_t hdr_n,
+                              const void* payload, size_t payload_n,
+                              bool backpressure) {
+  std::vector<char> frame(hdr_n + payload_n);
+  std::memcpy(frame.data(), hdr, hdr_n);
+  if (payload_n > 0)
+    std::memcpy(frame.data() + hdr_n, payload, payload_n);
+  {
+    std::unique_lock<std::mutex> lk(out_m_);
+    if (backpressure) {
+      out_cv_.wait(lk, [&] {
+        return dead_.load() || !running_.load() || out_bytes_ < kMaxQueuedBytes;
+      });
+      TMA_CHECK(!dead_.load() && running_.load(),
+                "parameter server event loop is not running");
+    }
+    out_bytes_ += frame.size();
+    outq_[peer].frames.push_back(std::move(frame));
+  }
+  if (wake_pipe_[1] >= 0) {
+    char b = 1;
+    (void)!write(wake_pipe_[1], &b, 1);
+  }
+}
+
+bool PsContext::drain_out(int peer) {
+  int fd = mesh_fd(peer);
+  while (true) {
+    std::vector<char>* front;
+    size_t off;
+    {
+      std::lock_guard<std::mutex> lk(out_m_);
+      auto& q = outq_[peer];
+      if (q.frames.empty()) return true;
+      front = &q.frames.front();
+      off = q.head_off;
+    }
+    // Only this thread pops/writes, so the front pointer stays valid
+    // outside the lock; enqueuers only push_back.
+    ssize_t n = ::send(fd, front->data() + off, front->size() - off, MSG_NOSIGNAL);
+    if (n < 0) {
+      if (errno == EINTR) continue;
+      if (errno == EAGAIN || errno == EWOULDBLOCK) return true;  // socket full
+      return false;
+    }
+    std::lock_guard<std::mutex> lk(out_m_);
+    auto& q = outq_[peer];
+    q.head_off += static_cast<size_t>(n);
+    out_bytes_ -= static_cast<size_t>(n);
+    if (q.head_off == q.frames.front().size()) {
+      q.frames.pop_front();
+      q.head_off = 0;
+    }
+    out_cv_.notify_all();
+  }
 }
 
 PsContext::Reg& PsContext::reg(int tid) {
@@ -86,13 +163,6 @@ void PsContext::shard_of(int tid, int r, size_t& off, size_t& cnt) {
   cnt = g.cnt[r];
 }
 
-void PsContext::write_frame(int peer, const void* hdr, size_t hdr_n,
-                            const void* payload, size_t payload_n) {
-  std::lock_guard<std::mutex> lk(*sock_m_[peer]);
-  mesh_.send_all(peer, hdr, hdr_n);
-  if (payload_n > 0) mesh_.send_all(peer, payload, payload_n);
-}
-
 std::shared_ptr<Work> PsContext::send(int tid, const void* data, int rule) {
   Reg& g = reg(tid);
   size_t es = dtype_size(g.dtype);
@@ -123,8 +193,9 @@ std::shared_ptr<Work> PsContext::send(int tid, const void* data, int rule) {
     FrameHdr h{kUpdate, static_cast<uint8_t>(rule),
                static_cast<uint8_t>(static_cast<int>(g.dtype)), 0, tid, seq,
                g.cnt[r] * es};
-    write_frame(r, &h, sizeof(h), static_cast<const char*>(data) + g.off[r] * es,
-                g.cnt[r] * es);
+    enqueue_frame(r, &h, sizeof(h),
+                  static_cast<const char*>(data) + g.off[r] * es,
+                  g.cnt[r] * es, /*backpressure=*/true);
   }
   if (remotes == 0) pend->work->finish(nullptr);
   return pend->work;
@@ -154,7 +225,7 @@ std::shared_ptr<Work> PsContext::receive(int tid, void* data) {
   for (int r = 0; r < world_; ++r) {
     if (r == rank_ || g.cnt[r] == 0) continue;
     FrameHdr h{kFetch, 0, 0, 0, tid, seq, 0};
-    write_frame(r, &h, sizeof(h), nullptr, 0);
+    enqueue_frame(r, &h, sizeof(h), nullptr, 0, /*backpressure=*/true);
   }
   if (remotes == 0) pend->work->finish(nullptr);
   return pend->work;
@@ -177,7 +248,7 @@ void PsContext::handle_frame(int peer) {
           cpu_reduce(g.shard.data(), buf.data(), g.cnt[rank_], g.dtype, RedOp::SUM);
       }
       FrameHdr ack{kAck, 0, 0, 0, h.tid, h.seq, 0};
-      write_frame(peer, &ack, sizeof(ack), nullptr, 0);
+      enqueue_frame(peer, &ack, sizeof(ack), nullptr, 0, /*backpressure=*/false);
       break;
     }
     case kFetch: {
@@ -189,7 +260,8 @@ void PsContext::handle_frame(int peer) {
       }
       FrameHdr d{kData, 0, static_cast<uint8_t>(static_cast<int>(g.dtype)), 0, h.tid,
                  h.seq, snapshot.size()};
-      write_frame(peer, &d, sizeof(d), snapshot.data(), snapshot.size());
+      enqueue_frame(peer, &d, sizeof(d), snapshot.data(), snapshot.size(),
+                    /*backpressure=*/false);
       break;
     }
     case kAck: {
@@ -243,32 +315,41 @@ void PsContext::loop() {
     peer_of.push_back(-1);
     for (int r = 0; r < world_; ++r) {
       if (r == rank_) continue;
-      // TcpMesh keeps one fd per peer; poll it for incoming frames.
-      fds.push_back({-1, POLLIN, 0});
+      // TcpMesh keeps one fd per peer; poll it for incoming frames, and
+      // for writability whenever its out-queue holds bytes. This thread is
+      // the mesh's only reader AND only writer (single-writer discipline),
+      // so it can never block mid-frame: a stalled peer costs queue memory,
+      // not liveness.
+      short ev = POLLIN;
+      {
+        std::lock_guard<std::mutex> lk(out_m_);
+        if (!outq_[r].frames.empty()) ev |= POLLOUT;
+      }
+      fds.push_back({mesh_fd(r), ev, 0});
       peer_of.push_back(r);
     }
-    // Fill fds from the mesh (private access via recv path: we poll the
-    // same sockets; safe because this thread is the only reader).
-    for (size_t i = 1; i < fds.size(); ++i) fds[i].fd = mesh_fd(peer_of[i]);
     int rc = poll(fds.data(), fds.size(), 500);
-    if (rc <= 0) continue;  // timeout tick or EINTR: re-check running_
+    if (rc < 0) continue;  // EINTR: re-check running_
     if (fds[0].revents & POLLIN) {
-      char b;
-      (void)!read(wake_pipe_[0], &b, 1);
-      continue;
+      char b[64];
+      (void)!read(wake_pipe_[0], b, sizeof(b));
     }
     for (size_t i = 1; i < fds.size(); ++i) {
-      if (fds[i].revents & POLLIN) {
-        try {
-          handle_frame(peer_of[i]);
-        } catch (const std::exception& e) {
-          if (!running_) return;
-          // Never std::terminate the process from the event loop: log and
-          // stop serving (pending client Works will time out loudly).
-          fprintf(stderr, "[torchmpi_amd ps] event loop error: %s\n", e.what());
-          running_ = false;
-          return;
+      try {
+        if (fds[i].revents & POLLOUT) {
+          if (!drain_out(peer_of[i]))
+            throw std::runtime_error("socket write error to peer " +
+                                     std::to_string(peer_of[i]));
         }
+        if (fds[i].revents & POLLIN) handle_frame(peer_of[i]);
+      } catch (const std::exception& e) {
+        if (!running_) return;
+        // Never std::terminate the process from the event loop: log, stop
+        // serving, and fail every pending Work so wait() raises loudly.
+        fprintf(stderr, "[torchmpi_amd ps] event loop error: %s\n", e.what());
+        running_ = false;
+        fail_all(std::string("event loop died: ") + e.what());
+        return;
       }
     }
   }

@@ -10,11 +10,21 @@
 // collective mesh so PS traffic never interleaves with collectives). One
 // event-loop thread per rank reads all peer sockets and dispatches frames:
 // requests are served against the local shard store; replies complete
-// pending client Works. Frame writes are serialized per socket by a mutex.
+// pending client Works.
+//
+// Single-writer discipline: ALL outbound frames (client kUpdate/kFetch and
+// loop-side kAck/kData) go through per-peer byte queues drained by the
+// event loop with non-blocking writes under POLLOUT. The loop thread never
+// blocks on a send, so two ranks simultaneously fetching each other's
+// large shards cannot mutually stall (each loop keeps reading while its
+// replies drain). Client enqueues are bounded (backpressure) and fail
+// loudly if the loop has died, so Work.wait() raises instead of hanging.
 // GPU tensors are staged through host memory by the Python wrapper (v1).
 #pragma once
 
 #include <atomic>
+#include <condition_variable>
+#include <deque>
 #include <memory>
 #include <mutex>
 #include <thread>
@@ -71,8 +81,17 @@ class PsContext {
   void loop();
   void handle_frame(int peer);
   int mesh_fd(int peer) { return mesh_.fd(peer); }
-  void write_frame(int peer, const void* hdr, size_t hdr_n, const void* payload,
-                   size_t payload_n);
+  // Queue one frame (header + payload copied) for the loop thread to write.
+  // backpressure=true (client threads) blocks while the queues hold more
+  // than kMaxQueuedBytes; false (the loop's own replies) never blocks.
+  void enqueue_frame(int peer, const void* hdr, size_t hdr_n, const void* payload,
+                     size_t payload_n, bool backpressure);
+  // Drain what the socket will take without blocking; returns false on a
+  // fatal socket error.
+  bool drain_out(int peer);
+  // Mark the loop dead: fail every pending Work so wait() raises, wake
+  // backpressure waiters, make further enqueues throw.
+  void fail_all(const std::string& why);
   Reg& reg(int tid);
 
   int rank_ = 0;
@@ -80,6 +99,7 @@ class PsContext {
   TcpMesh mesh_;
   std::thread loop_thread_;
   std::atomic<bool> running_{false};
+  std::atomic<bool> dead_{false};
   int wake_pipe_[2] = {-1, -1};
 
   std::mutex regs_m_;
@@ -89,7 +109,15 @@ class PsContext {
   std::unordered_map<uint64_t, std::shared_ptr<Pending>> pending_;
   std::atomic<uint64_t> next_seq_{1};
 
-  std::vector<std::unique_ptr<std::mutex>> sock_m_;  // per-peer write locks
+  static constexpr size_t kMaxQueuedBytes = size_t(64) << 20;
+  struct OutQ {
+    std::deque<std::vector<char>> frames;
+    size_t head_off = 0;  // bytes of frames.front() already written
+  };
+  std::mutex out_m_;
+  std::condition_variable out_cv_;
+  std::vector<OutQ> outq_;       // [peer]
+  size_t out_bytes_ = 0;         // total queued across peers
 };
 
 }  // namespace tma

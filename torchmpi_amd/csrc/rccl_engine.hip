@@ -487,43 +487,66 @@ class RcclEngine {
       return work;
     }
     hipEvent_t ready = record_ready();
-    auto s = streams_[0];
-    TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
-    order_in(s);
+    auto s0 = streams_[0];
+    TMA_HIP_CHECK(hipStreamWaitEvent(s0.stream(), ready, 0));
+    order_in(s0);
 
     // Non-root ranks work on a scratch copy so their input is preserved
-    // (MPI_Reduce semantics, reference lib/collectives.cpp reduce).
-    char* base = static_cast<char*>(t.data_ptr());
+    // (MPI_Reduce semantics, reference lib/collectives.cpp reduce). The
+    // root reduces in place: its tensor becomes the result anyway.
+    char* tbase = static_cast<char*>(t.data_ptr());
+    char* base = tbase;
     at::Tensor scratch;
     if (rank_ != root) {
-      scratch = scratch_like(count * es, s);
+      scratch = scratch_like(count * es, s0);
       work->keep.push_back(scratch);
-      TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), ready, 0));
-      TMA_HIP_CHECK(hipMemcpyAsync(scratch.data_ptr(), base, count * es,
-                                   hipMemcpyDeviceToDevice, s.stream()));
+      TMA_HIP_CHECK(hipMemcpyAsync(scratch.data_ptr(), tbase, count * es,
+                                   hipMemcpyDeviceToDevice, s0.stream()));
       base = static_cast<char*>(scratch.data_ptr());
     }
-    ring_reduce_scatter(0, base, count, d, op, work);
-    // Gather segments to root: rank r owns segment r after reduce-scatter.
-    std::vector<size_t> off, cnt;
-    split_segments(count, world_, off, cnt);
-    char* tbase = static_cast<char*>(t.data_ptr());
-    TMA_NCCL_CHECK(ncclGroupStart());
-    if (rank_ == root) {
-      for (int r = 0; r < world_; ++r) {
-        if (r == root || cnt[r] == 0) continue;
-        TMA_NCCL_CHECK(ncclRecv(tbase + off[r] * es, cnt[r] * es, ncclInt8, r,
-                                comms_[0], s.stream()));
-      }
-    } else {
-      if (cnt[rank_] > 0)
-        TMA_NCCL_CHECK(ncclSend(base + off[rank_] * es, cnt[rank_] * es, ncclInt8,
-                                root, comms_[0], s.stream()));
+
+    // Multi-ring (same striping as allreduce): ring r reduce-scatters its
+    // stripe then gathers the stripe's segments to root over comms_[r],
+    // using all xGMI links instead of one.
+    const size_t align = 16 / es;
+    int nrings = count * es >= multiring_min_bytes_ ? used_rings() : 1;
+    std::vector<size_t> soff, scnt;
+    split_aligned(count, nrings, align, soff, scnt);
+    hipEvent_t fork = nullptr;
+    if (nrings > 1) {
+      TMA_HIP_CHECK(hipEventCreateWithFlags(&fork, hipEventDisableTiming));
+      TMA_HIP_CHECK(hipEventRecord(fork, s0.stream()));
     }
-    TMA_NCCL_CHECK(ncclGroupEnd());
-    record_stream_uses(t, s);
-    record_done(work, s);
-    order_out(1);
+    for (int r = 0; r < nrings; ++r) {
+      if (scnt[r] == 0) continue;
+      auto s = streams_[r];
+      if (r > 0) {
+        TMA_HIP_CHECK(hipStreamWaitEvent(s.stream(), fork, 0));
+        order_in(s);
+      }
+      ring_reduce_scatter(r, base + soff[r] * es, scnt[r], d, op, work);
+      // Gather stripe segments to root: after RS, rank g owns segment g.
+      std::vector<size_t> off, cnt;
+      split_segments(scnt[r], world_, off, cnt);
+      TMA_NCCL_CHECK(ncclGroupStart());
+      if (rank_ == root) {
+        for (int g = 0; g < world_; ++g) {
+          if (g == root || cnt[g] == 0) continue;
+          TMA_NCCL_CHECK(ncclRecv(tbase + (soff[r] + off[g]) * es, cnt[g] * es,
+                                  ncclInt8, g, comms_[r], s.stream()));
+        }
+      } else {
+        if (cnt[rank_] > 0)
+          TMA_NCCL_CHECK(ncclSend(base + (soff[r] + off[rank_]) * es,
+                                  cnt[rank_] * es, ncclInt8, root, comms_[r],
+                                  s.stream()));
+      }
+      TMA_NCCL_CHECK(ncclGroupEnd());
+      record_stream_uses(t, s);
+      record_done(work, s);
+    }
+    order_out(nrings);
+    if (fork) TMA_HIP_CHECK(hipEventDestroy(fork));
     TMA_HIP_CHECK(hipEventDestroy(ready));
     return work;
   }
@@ -745,7 +768,7 @@ class RcclEngine {
  private:
   // Record an event on the caller's current stream (orders the collective
   // after e.g. the backward kernels that produced the gradient — reference
-  // stack 3.3's cudaEventRecord(ev_ready, currentStream)).
+  // stack 3.3's ev_ready event record on currentStream).
   hipEvent_t record_ready() {
     hipEvent_t ev;
     TMA_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));

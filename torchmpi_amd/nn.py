@@ -103,6 +103,7 @@ class GradientSynchronizer:
                 self._param_bucket[id(p)] = bi
 
         self._hooks = []
+        self._accum_mode = False
         if _register_hooks and self.comm.size() > 1:
             for p in params:
                 h = p.register_post_accumulate_grad_hook(self._hook)
@@ -129,12 +130,37 @@ class GradientSynchronizer:
     # guarantee; DDP enforces the same rule).
 
     def _hook(self, p):
+        if self._accum_mode:
+            return  # no_sync(): grads accumulate locally, no comm
         bi = self._param_bucket[id(p)]
         b = self.buckets[bi]
+        if b["pending"] <= 0:
+            raise RuntimeError(
+                "GradientSynchronizer: a gradient hook fired on an "
+                "already-reduced bucket — multiple backward() calls before "
+                "finish() would silently overwrite the accumulated grads "
+                "with a stale reduction. Wrap the non-final backward passes "
+                "in sync.no_sync() for gradient accumulation.")
         b["pending"] -= 1
         if b["pending"] == 0:
             b["ready"] = True
             self._launch_ready_in_order()
+
+    def no_sync(self):
+        """Context manager: skip gradient allreduce inside (DDP-style
+        gradient accumulation). The final backward runs outside no_sync()
+        and reduces the accumulated grads."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def ctx():
+            prev = self._accum_mode
+            self._accum_mode = True
+            try:
+                yield
+            finally:
+                self._accum_mode = prev
+        return ctx()
 
     def _launch_ready_in_order(self):
         while (self._next_launch < len(self.buckets)

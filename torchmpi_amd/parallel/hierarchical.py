@@ -19,7 +19,7 @@ def hierarchical_allreduce(tensor: torch.Tensor, intra, inter, average=False):
     in exactly one intra group and one inter group, and inter groups connect
     ranks holding the same shard index.
     """
-    from .. import allgather, allreduce, reduce_scatter
+    from .. import allgather, allreduce, reduce_scatter_
 
     wi = intra.size()
     n = tensor.numel()
@@ -28,8 +28,14 @@ def hierarchical_allreduce(tensor: torch.Tensor, intra, inter, average=False):
     flat = tensor.view(-1)
     shard = torch.empty(my_count, dtype=tensor.dtype, device=tensor.device)
 
-    reduce_scatter(flat, shard, comm=intra)
-    allreduce(shard, comm=inter, average=average)
+    # In-place variant: flat is rewritten by the allgather below anyway,
+    # so the copy the preserving reduce_scatter makes would be wasted.
+    reduce_scatter_(flat, shard, comm=intra)
+    # The value is summed over all wi*we ranks, so the mean divides by the
+    # product — averaging only at the inter level would leave it wi x too big.
+    allreduce(shard, comm=inter)
+    if average:
+        shard /= float(wi * inter.size())
 
     if rem == 0:
         # equal shards: allgather straight back into the tensor
