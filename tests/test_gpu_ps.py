@@ -49,6 +49,29 @@ def _ps_worker(rank, world, port, scenario):
         tm.barrier()
         ps.receive(p, out=got).wait()
         assert torch.equal(got, torch.arange(n, dtype=torch.float32, device=dev))
+    elif scenario == "rules16":
+        # bf16 'add' rule (32-bit system-scope CAS path) with an ODD count:
+        # exercises the tail-word CAS and non-16B shard offsets.
+        n = 1031
+        p = torch.zeros(n, device=dev, dtype=torch.bfloat16)
+        ps.register(p)
+        tm.barrier()
+        ps.send(p, "add",
+                data=torch.full((n,), float(rank + 1), device=dev,
+                                dtype=torch.bfloat16)).wait()
+        tm.barrier()
+        got = torch.empty(n, device=dev, dtype=torch.bfloat16)
+        ps.receive(p, out=got).wait()
+        expect = float(W * (W + 1) / 2)
+        assert torch.allclose(got.float(), torch.full((n,), expect, device=dev),
+                              rtol=1e-2), got[:5]
+        tm.barrier()
+        if rank == 0:
+            ps.send(p, "copy",
+                    data=torch.arange(n, device=dev).to(torch.bfloat16)).wait()
+        tm.barrier()
+        ps.receive(p, out=got).wait()
+        assert torch.equal(got, torch.arange(n, device=dev).to(torch.bfloat16))
     elif scenario == "shards":
         p = torch.full((64,), float(rank), device=dev)
         ps.register(p)  # shard k comes from rank k's initial value
@@ -65,7 +88,7 @@ def _ps_worker(rank, world, port, scenario):
 
 
 @requires_gpu
-@pytest.mark.parametrize("scenario", ["rules", "shards"])
+@pytest.mark.parametrize("scenario", ["rules", "rules16", "shards"])
 def test_gpu_ps_two_procs_one_device(scenario):
     import torch.multiprocessing as mp
     port = find_free_port()

@@ -33,6 +33,32 @@ def test_ps_rules():
     run_distributed(_rules_body, world=4)
 
 
+def _mutual_large_fetch_body(tm):
+    """Regression (advisor, round 1): two ranks simultaneously fetching each
+    other's shards LARGER than the combined socket buffers used to mutually
+    block both event loops (blocking kData sends). The loops now drain
+    outbound queues non-blockingly, so this completes."""
+    W, r = tm.size(), tm.rank()
+    n = 8 << 20  # 32 MB fp32: shard snapshots far exceed socket buffers
+    ps = tm.ParameterServer()
+    p = torch.full((n,), float(r + 1))
+    ps.register(p)
+    tm.barrier()
+    got = torch.empty(n)
+    ps.receive(p, out=got).wait()  # both ranks fetch at the same time
+    base = n // W
+    off = 0
+    for k in range(W):
+        cnt = base + (1 if k < n % W else 0)
+        assert torch.equal(got[off], torch.tensor(float(k + 1))), (k, got[off])
+        off += cnt
+    ps.shutdown()
+
+
+def test_ps_mutual_large_fetch():
+    run_distributed(_mutual_large_fetch_body, world=2)
+
+
 def _receive_into_registered_body(tm):
     W, r = tm.size(), tm.rank()
     ps = tm.ParameterServer()

@@ -37,6 +37,9 @@ void PsContext::start(int rank, const std::vector<std::string>& addrs) {
   if (world_ > 1) mesh_.connect_mesh(rank, addrs);
   outq_.clear();
   outq_.resize(world_);
+  in_.clear();
+  in_.resize(world_);
+  for (auto& st : in_) st.buf.resize(sizeof(FrameHdr));
   out_bytes_ = 0;
   dead_ = false;
   TMA_CHECK(pipe(wake_pipe_) == 0, "pipe failed");
@@ -231,21 +234,50 @@ std::shared_ptr<Work> PsContext::receive(int tid, void* data) {
   return pend->work;
 }
 
-void PsContext::handle_frame(int peer) {
+bool PsContext::read_in(int peer) {
+  int fd = mesh_fd(peer);
+  auto& st = in_[peer];
+  while (true) {
+    ssize_t n = ::recv(fd, st.buf.data() + st.filled,
+                       st.buf.size() - st.filled, MSG_DONTWAIT);
+    if (n == 0) return false;  // peer closed
+    if (n < 0) {
+      if (errno == EINTR) continue;
+      if (errno == EAGAIN || errno == EWOULDBLOCK) return true;
+      return false;
+    }
+    st.filled += static_cast<size_t>(n);
+    if (st.filled < st.buf.size()) continue;
+    if (!st.have_hdr) {
+      FrameHdr h;
+      std::memcpy(&h, st.buf.data(), sizeof(h));
+      st.have_hdr = true;
+      if (h.bytes > 0) {
+        st.buf.resize(sizeof(FrameHdr) + h.bytes);
+        std::memcpy(st.buf.data(), &h, sizeof(h));
+        continue;  // keep filling until the payload is complete
+      }
+    }
+    dispatch(peer, st.buf.data(), st.buf.data() + sizeof(FrameHdr));
+    st.buf.assign(sizeof(FrameHdr), 0);
+    st.filled = 0;
+    st.have_hdr = false;
+  }
+}
+
+void PsContext::dispatch(int peer, const void* hdr, const char* payload) {
   FrameHdr h;
-  mesh_.recv_all(peer, &h, sizeof(h));
+  std::memcpy(&h, hdr, sizeof(h));
   switch (h.type) {
     case kUpdate: {
       Reg& g = reg(h.tid);
-      std::vector<char> buf(h.bytes);
-      mesh_.recv_all(peer, buf.data(), h.bytes);
       {
         std::lock_guard<std::mutex> lk(g.m);
         TMA_CHECK(h.bytes == g.shard.size(), "update size mismatch");
         if (h.rule == 0)
-          std::memcpy(g.shard.data(), buf.data(), h.bytes);
+          std::memcpy(g.shard.data(), payload, h.bytes);
         else
-          cpu_reduce(g.shard.data(), buf.data(), g.cnt[rank_], g.dtype, RedOp::SUM);
+          cpu_reduce(g.shard.data(), payload, g.cnt[rank_], g.dtype, RedOp::SUM);
       }
       FrameHdr ack{kAck, 0, 0, 0, h.tid, h.seq, 0};
       enqueue_frame(peer, &ack, sizeof(ack), nullptr, 0, /*backpressure=*/false);
@@ -290,7 +322,9 @@ void PsContext::handle_frame(int peer) {
       TMA_CHECK(p != nullptr, "DATA reply with no pending receive");
       Reg& g = reg(h.tid);
       size_t es = dtype_size(g.dtype);
-      mesh_.recv_all(peer, static_cast<char*>(p->dest) + g.off[peer] * es, h.bytes);
+      TMA_CHECK(h.bytes == g.cnt[peer] * es, "DATA size mismatch");
+      std::memcpy(static_cast<char*>(p->dest) + g.off[peer] * es, payload,
+                  h.bytes);
       if (p->remaining.fetch_sub(1) == 1) {
         {
           std::lock_guard<std::mutex> lk(pend_m_);
@@ -341,7 +375,11 @@ void PsContext::loop() {
             throw std::runtime_error("socket write error to peer " +
                                      std::to_string(peer_of[i]));
         }
-        if (fds[i].revents & POLLIN) handle_frame(peer_of[i]);
+        if (fds[i].revents & (POLLIN | POLLHUP | POLLERR)) {
+          if (!read_in(peer_of[i]))
+            throw std::runtime_error("socket read error/close from peer " +
+                                     std::to_string(peer_of[i]));
+        }
       } catch (const std::exception& e) {
         if (!running_) return;
         // Never std::terminate the process from the event loop: log, stop

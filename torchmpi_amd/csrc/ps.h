@@ -79,7 +79,13 @@ class PsContext {
   };
 
   void loop();
-  void handle_frame(int peer);
+  // Non-blocking read pump: accumulates bytes into the peer's frame buffer
+  // and dispatches every complete frame; returns false on a fatal socket
+  // error. The loop thread must never block on a read — a 16 MB kData
+  // payload dribbles in across many poll iterations while THIS rank's own
+  // replies keep draining (the round-1 advisor deadlock, both directions).
+  bool read_in(int peer);
+  void dispatch(int peer, const void* hdr, const char* payload);
   int mesh_fd(int peer) { return mesh_.fd(peer); }
   // Queue one frame (header + payload copied) for the loop thread to write.
   // backpressure=true (client threads) blocks while the queues hold more
@@ -118,6 +124,13 @@ class PsContext {
   std::condition_variable out_cv_;
   std::vector<OutQ> outq_;       // [peer]
   size_t out_bytes_ = 0;         // total queued across peers
+
+  struct InSt {
+    std::vector<char> buf;  // header, then header+payload
+    size_t filled = 0;
+    bool have_hdr = false;
+  };
+  std::vector<InSt> in_;         // [peer], loop thread only
 };
 
 }  // namespace tma
