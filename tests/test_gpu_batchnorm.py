@@ -52,6 +52,79 @@ def test_bn_forward_backward_matches_torch(dtype, shape):
 
 
 @requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("with_res", [False, True])
+def test_bn_fused_relu_residual(dtype, with_res):
+    """y = relu(bn(x) [+ res]) fused in the BN epilogue vs the unfused
+    torch fp32 reference, forward + backward incl. the residual grad."""
+    from torchmpi_amd.ops.batchnorm import FusedBatchNorm2d
+    torch.manual_seed(3)
+    N, C, H, W = 4, 64, 14, 14
+    dev = torch.device("cuda", 0)
+    ref_bn = torch.nn.BatchNorm2d(C).to(dev)
+    our_bn = FusedBatchNorm2d.from_batchnorm(ref_bn).to(dev)
+
+    x = torch.randn(N, C, H, W, device=dev)
+    res = torch.randn(N, C, H, W, device=dev)
+    x_ref = x.clone().requires_grad_(True)
+    r_ref = res.clone().requires_grad_(True)
+    x_our = x.clone().to(dtype).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    r_our = res.clone().to(dtype).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+
+    y_ref = ref_bn(x_ref)
+    if with_res:
+        y_ref = y_ref + r_ref
+    y_ref = torch.relu(y_ref)
+    y_our = our_bn(x_our, relu=True, res=r_our if with_res else None)
+    torch.cuda.synchronize()
+    tol = 5e-2 if dtype == torch.bfloat16 else 2e-5
+    assert torch.allclose(y_our.float(), y_ref, atol=tol, rtol=tol), \
+        (y_our.float() - y_ref).abs().max()
+
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y_our.backward(g.to(dtype).contiguous(memory_format=torch.channels_last))
+    torch.cuda.synchronize()
+    assert torch.allclose(x_our.grad.float(), x_ref.grad, atol=10 * tol,
+                          rtol=10 * tol), \
+        (x_our.grad.float() - x_ref.grad).abs().max()
+    if with_res:
+        assert torch.allclose(r_our.grad.float(), r_ref.grad, atol=10 * tol,
+                              rtol=10 * tol), \
+            (r_our.grad.float() - r_ref.grad).abs().max()
+    assert torch.allclose(our_bn.weight.grad, ref_bn.weight.grad,
+                          atol=10 * tol, rtol=10 * tol)
+    assert torch.allclose(our_bn.bias.grad, ref_bn.bias.grad, atol=10 * tol,
+                          rtol=10 * tol)
+
+
+@requires_gpu
+def test_bn_fused_relu_odd_channels():
+    """Fused relu on the scalar-fallback path (C not divisible by VEC)."""
+    from torchmpi_amd.ops.batchnorm import FusedBatchNorm2d
+    torch.manual_seed(4)
+    N, C, H, W = 3, 67, 9, 9
+    dev = torch.device("cuda", 0)
+    ref_bn = torch.nn.BatchNorm2d(C).to(dev)
+    our_bn = FusedBatchNorm2d.from_batchnorm(ref_bn).to(dev)
+    x = torch.randn(N, C, H, W, device=dev)
+    x_ref = x.clone().requires_grad_(True)
+    x_our = x.clone().contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    y_ref = torch.relu(ref_bn(x_ref))
+    y_our = our_bn(x_our, relu=True)
+    torch.cuda.synchronize()
+    assert torch.allclose(y_our.float(), y_ref, atol=2e-5, rtol=2e-5)
+    g = torch.randn_like(y_ref)
+    y_ref.backward(g)
+    y_our.backward(g.contiguous(memory_format=torch.channels_last))
+    torch.cuda.synchronize()
+    assert torch.allclose(x_our.grad, x_ref.grad, atol=2e-4, rtol=2e-4)
+
+
+@requires_gpu
 def test_bn_eval_mode():
     from torchmpi_amd.ops.batchnorm import FusedBatchNorm2d
     torch.manual_seed(1)

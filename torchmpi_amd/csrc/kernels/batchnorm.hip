@@ -249,13 +249,17 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
 }
 
 // ---- forward apply (fixed channels per thread, params in registers) ------
+// RELU / RES fuse the following activation and residual add into the BN
+// epilogue: y = [relu](sc*x + sh [+ res]) — saves the separate elementwise
+// passes that were ~20% of the ResNet-50 step (profiles/
+// resnet50_bf16_kernel_stats.md).
 
-template <typename T, int VEC>
+template <typename T, int VEC, bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_fwd_apply_vec(
-    const T* __restrict__ x, T* __restrict__ y, const float* __restrict__ mean,
-    const float* __restrict__ invstd, const float* __restrict__ weight,
-    const float* __restrict__ bias, size_t M, int nC8, int c8_per_block,
-    int row_blocks) {
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ weight, const float* __restrict__ bias, size_t M,
+    int nC8, int c8_per_block, int row_blocks) {
   const int c_blocks = nC8 / c8_per_block;
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
@@ -273,39 +277,53 @@ __global__ __launch_bounds__(256) void bn_fwd_apply_vec(
   }
   const size_t row_stride = (size_t)row_blocks * rows_per_iter;
   const T* xb = x + (size_t)c8 * VEC;
+  const T* rb2 = RES ? res + (size_t)c8 * VEC : nullptr;
   T* yb = y + (size_t)c8 * VEC;
   for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
-    T v[VEC], o[VEC];
+    T v[VEC], o[VEC], rv[VEC];
     *reinterpret_cast<uint4*>(v) =
         *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
+    if (RES)
+      *reinterpret_cast<uint4*>(rv) =
+          *reinterpret_cast<const uint4*>(rb2 + r * (size_t)C);
 #pragma unroll
-    for (int e = 0; e < VEC; ++e) o[e] = stf<T>(ldf(v[e]) * sc[e] + sh[e]);
+    for (int e = 0; e < VEC; ++e) {
+      float f = ldf(v[e]) * sc[e] + sh[e];
+      if (RES) f += ldf(rv[e]);
+      if (RELU) f = fmaxf(f, 0.f);
+      o[e] = stf<T>(f);
+    }
     *reinterpret_cast<uint4*>(yb + r * (size_t)C) = *reinterpret_cast<uint4*>(o);
   }
 }
 
-template <typename T>
+template <typename T, bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_fwd_apply_scalar(
-    const T* __restrict__ x, T* __restrict__ y, const float* __restrict__ mean,
-    const float* __restrict__ invstd, const float* __restrict__ weight,
-    const float* __restrict__ bias, size_t total, int C) {
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ weight, const float* __restrict__ bias,
+    size_t total, int C) {
   const size_t stride = (size_t)gridDim.x * blockDim.x;
   for (size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
     int c = (int)(i % C);
     float xh = (ldf(x[i]) - mean[c]) * invstd[c];
-    y[i] = stf<T>(xh * weight[c] + bias[c]);
+    float f = xh * weight[c] + bias[c];
+    if (RES) f += ldf(res[i]);
+    if (RELU) f = fmaxf(f, 0.f);
+    y[i] = stf<T>(f);
   }
 }
 
 // ---- backward stats ------------------------------------------------------
 
-template <typename T, int VEC>
+template <typename T, int VEC, bool RELU>
 __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
     const T* __restrict__ x, const T* __restrict__ dy,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ psdy, float* __restrict__ psdyx, size_t M, int nC8,
-    int c8_per_block, int row_blocks) {
+    const T* __restrict__ yout, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ psdy,
+    float* __restrict__ psdyx, size_t M, int nC8, int c8_per_block,
+    int row_blocks) {
   const int c_blocks = nC8 / c8_per_block;
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
@@ -324,15 +342,20 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
   const size_t row_stride = (size_t)row_blocks * rows_per_iter;
   const T* xb = x + (size_t)c8 * VEC;
   const T* gb = dy + (size_t)c8 * VEC;
+  const T* yb = RELU ? yout + (size_t)c8 * VEC : nullptr;
   for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
-    T xv[VEC], gv[VEC];
+    T xv[VEC], gv[VEC], yv[VEC];
     *reinterpret_cast<uint4*>(xv) =
         *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
     *reinterpret_cast<uint4*>(gv) =
         *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
+    if (RELU)
+      *reinterpret_cast<uint4*>(yv) =
+          *reinterpret_cast<const uint4*>(yb + r * (size_t)C);
 #pragma unroll
     for (int e = 0; e < VEC; ++e) {
       float g = ldf(gv[e]);
+      if (RELU && ldf(yv[e]) <= 0.f) g = 0.f;  // relu mask from fused output
       float xh = (ldf(xv[e]) - mu[e]) * is[e];
       acc[e] += g;
       accx[e] += g * xh;
@@ -369,12 +392,12 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
   }
 }
 
-template <typename T>
+template <typename T, bool RELU>
 __global__ __launch_bounds__(256) void bn_bwd_stats_scalar_kernel(
     const T* __restrict__ x, const T* __restrict__ dy,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    float* __restrict__ sdy, float* __restrict__ sdyx, size_t M, int C,
-    int c_blocks, int row_blocks) {
+    const T* __restrict__ yout, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ sdy,
+    float* __restrict__ sdyx, size_t M, int C, int c_blocks, int row_blocks) {
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
   const int lane = (int)threadIdx.x % 64;
@@ -386,6 +409,7 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_scalar_kernel(
     const size_t row_stride = (size_t)row_blocks * 4;
     for (size_t r = (size_t)rb * 4 + rsub; r < M; r += row_stride) {
       float g = ldf(dy[r * C + c]);
+      if (RELU && ldf(yout[r * C + c]) <= 0.f) g = 0.f;
       float xh = (ldf(x[r * C + c]) - mu) * is;
       s += g;
       sx += g * xh;
@@ -410,9 +434,12 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_scalar_kernel(
 
 // ---- backward apply ------------------------------------------------------
 
-template <typename T, int VEC>
+// RELU: dy is masked by (yout > 0) everywhere it appears; RES: the residual
+// branch's gradient dres = masked dy is produced in the same pass.
+template <typename T, int VEC, bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_bwd_apply_vec(
-    const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
+    const T* __restrict__ x, const T* __restrict__ dy,
+    const T* __restrict__ yout, T* __restrict__ dx, T* __restrict__ dres,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ sdy,
     const float* __restrict__ sdyx, float invM, size_t M, int nC8,
@@ -425,7 +452,7 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_vec(
   const int rsub = (int)threadIdx.x / c8_per_block;
   const int c8 = cb * c8_per_block + c8l;
   const int C = nC8 * VEC;
-  // dx = w*is*(dy - sdy/M - xhat*sdyx/M) = A*dy + B*x + D
+  // dx = w*is*(g - sdy/M - xhat*sdyx/M) = A*g + B*x + D  (g = masked dy)
   float A[VEC], B[VEC], D[VEC];
 #pragma unroll
   for (int e = 0; e < VEC; ++e) {
@@ -439,23 +466,36 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_vec(
   const size_t row_stride = (size_t)row_blocks * rows_per_iter;
   const T* xb = x + (size_t)c8 * VEC;
   const T* gb = dy + (size_t)c8 * VEC;
+  const T* yb = RELU ? yout + (size_t)c8 * VEC : nullptr;
   T* ob = dx + (size_t)c8 * VEC;
+  T* eb = RES ? dres + (size_t)c8 * VEC : nullptr;
   for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
-    T xv[VEC], gv[VEC], o[VEC];
+    T xv[VEC], gv[VEC], yv[VEC], o[VEC], ev[VEC];
     *reinterpret_cast<uint4*>(xv) =
         *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
     *reinterpret_cast<uint4*>(gv) =
         *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
+    if (RELU)
+      *reinterpret_cast<uint4*>(yv) =
+          *reinterpret_cast<const uint4*>(yb + r * (size_t)C);
 #pragma unroll
-    for (int e = 0; e < VEC; ++e)
-      o[e] = stf<T>(A[e] * ldf(gv[e]) + B[e] * ldf(xv[e]) + D[e]);
+    for (int e = 0; e < VEC; ++e) {
+      float g = ldf(gv[e]);
+      if (RELU && ldf(yv[e]) <= 0.f) g = 0.f;
+      o[e] = stf<T>(A[e] * g + B[e] * ldf(xv[e]) + D[e]);
+      if (RES) ev[e] = stf<T>(g);
+    }
     *reinterpret_cast<uint4*>(ob + r * (size_t)C) = *reinterpret_cast<uint4*>(o);
+    if (RES)
+      *reinterpret_cast<uint4*>(eb + r * (size_t)C) =
+          *reinterpret_cast<uint4*>(ev);
   }
 }
 
-template <typename T>
+template <typename T, bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_bwd_apply_scalar(
-    const T* __restrict__ x, const T* __restrict__ dy, T* __restrict__ dx,
+    const T* __restrict__ x, const T* __restrict__ dy,
+    const T* __restrict__ yout, T* __restrict__ dx, T* __restrict__ dres,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ sdy,
     const float* __restrict__ sdyx, float invM, size_t total, int C) {
@@ -465,8 +505,10 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_scalar(
     int c = (int)(i % C);
     float xh = (ldf(x[i]) - mean[c]) * invstd[c];
     float g = ldf(dy[i]);
+    if (RELU && ldf(yout[i]) <= 0.f) g = 0.f;
     float t = g - sdy[c] * invM - xh * sdyx[c] * invM;
     dx[i] = stf<T>(weight[c] * invstd[c] * t);
+    if (RES) dres[i] = stf<T>(g);
   }
 }
 
@@ -485,11 +527,21 @@ int scalar_grid(size_t work) {
 
 }  // namespace
 
-// Returns (y, save_mean, save_invstd).
+// Returns (y, save_mean, save_invstd). relu/res fuse the activation and
+// residual add into the apply epilogue (y = [relu](bn(x) [+ res])).
 std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor bias,
                                    at::Tensor running_mean, at::Tensor running_var,
-                                   bool training, double momentum, double eps) {
+                                   bool training, double momentum, double eps,
+                                   bool relu, c10::optional<at::Tensor> res_opt) {
   check_bn_input(x);
+  at::Tensor res = res_opt.value_or(at::Tensor());
+  const bool has_res = res.defined();
+  if (has_res) {
+    TMA_CHECK(res.sizes() == x.sizes() && res.scalar_type() == x.scalar_type(),
+              "bn: residual shape/dtype mismatch");
+    TMA_CHECK(res.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "bn: residual must be channels_last");
+  }
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   const int C = (int)x.size(1);
   const size_t M = (size_t)x.numel() / C;
@@ -555,45 +607,59 @@ std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor b
   }
 
   at::Tensor y = at::empty_like(x);
-  if (vg.ok) {
-    if (bf16)
-      hipLaunchKernelGGL((bn_fwd_apply_vec<__hip_bfloat16, 8>), dim3(vg.grid),
-                         dim3(256), 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         (__hip_bfloat16*)y.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         bias.data_ptr<float>(), M, vg.nC8, vg.c8_per_block,
-                         vg.row_blocks);
-    else
-      hipLaunchKernelGGL((bn_fwd_apply_vec<float, 4>), dim3(vg.grid), dim3(256),
-                         0, stream, x.data_ptr<float>(), y.data_ptr<float>(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         weight.data_ptr<float>(), bias.data_ptr<float>(), M,
-                         vg.nC8, vg.c8_per_block, vg.row_blocks);
-  } else {
-    size_t total = (size_t)x.numel();
-    if (bf16)
-      hipLaunchKernelGGL(bn_fwd_apply_scalar<__hip_bfloat16>,
-                         dim3(scalar_grid(total)), dim3(256), 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         (__hip_bfloat16*)y.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         bias.data_ptr<float>(), total, C);
-    else
-      hipLaunchKernelGGL(bn_fwd_apply_scalar<float>, dim3(scalar_grid(total)),
-                         dim3(256), 0, stream, x.data_ptr<float>(),
-                         y.data_ptr<float>(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         bias.data_ptr<float>(), total, C);
-  }
+#define TMA_BN_FWD(T, VEC)                                                     \
+  do {                                                                         \
+    const T* xp = (const T*)x.data_ptr();                                      \
+    const T* rp = has_res ? (const T*)res.data_ptr() : nullptr;                \
+    T* yp = (T*)y.data_ptr();                                                  \
+    if (vg.ok) {                                                               \
+      auto args = [&](auto kern) {                                             \
+        hipLaunchKernelGGL(kern, dim3(vg.grid), dim3(256), 0, stream, xp, rp,  \
+                           yp, mean.data_ptr<float>(),                         \
+                           invstd.data_ptr<float>(), weight.data_ptr<float>(), \
+                           bias.data_ptr<float>(), M, vg.nC8, vg.c8_per_block, \
+                           vg.row_blocks);                                     \
+      };                                                                       \
+      if (relu && has_res) args(bn_fwd_apply_vec<T, VEC, true, true>);         \
+      else if (relu) args(bn_fwd_apply_vec<T, VEC, true, false>);              \
+      else if (has_res) args(bn_fwd_apply_vec<T, VEC, false, true>);           \
+      else args(bn_fwd_apply_vec<T, VEC, false, false>);                       \
+    } else {                                                                   \
+      size_t total = (size_t)x.numel();                                        \
+      auto args = [&](auto kern) {                                             \
+        hipLaunchKernelGGL(kern, dim3(scalar_grid(total)), dim3(256), 0,       \
+                           stream, xp, rp, yp, mean.data_ptr<float>(),         \
+                           invstd.data_ptr<float>(), weight.data_ptr<float>(), \
+                           bias.data_ptr<float>(), total, C);                  \
+      };                                                                       \
+      if (relu && has_res) args(bn_fwd_apply_scalar<T, true, true>);           \
+      else if (relu) args(bn_fwd_apply_scalar<T, true, false>);                \
+      else if (has_res) args(bn_fwd_apply_scalar<T, false, true>);             \
+      else args(bn_fwd_apply_scalar<T, false, false>);                         \
+    }                                                                          \
+  } while (0)
+  if (bf16)
+    TMA_BN_FWD(__hip_bfloat16, 8);
+  else
+    TMA_BN_FWD(float, 4);
+#undef TMA_BN_FWD
   return {y, mean, invstd};
 }
 
-// Returns (dx, dweight, dbias).
+// Returns (dx, dweight, dbias[, dres]). `yout` is the fused forward output
+// (needed for the relu mask); `need_dres` additionally emits the residual
+// branch's gradient (= relu-masked dy) in the same pass.
 std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weight,
-                                    at::Tensor mean, at::Tensor invstd) {
+                                    at::Tensor mean, at::Tensor invstd,
+                                    c10::optional<at::Tensor> yout_opt, bool relu,
+                                    bool need_dres) {
+  at::Tensor yout = yout_opt.value_or(at::Tensor());
   check_bn_input(x);
   check_bn_input(dy);
+  if (relu) {
+    TMA_CHECK(yout.defined(), "bn_backward: relu needs the forward output");
+    check_bn_input(yout);
+  }
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   const int C = (int)x.size(1);
   const size_t M = (size_t)x.numel() / C;
@@ -603,94 +669,93 @@ std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weig
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
   at::Tensor sdy = at::empty({C}, fopt);
   at::Tensor sdyx = at::empty({C}, fopt);
-
-  if (vg.ok) {
-    at::Tensor psdy = at::empty({(int64_t)vg.row_blocks * C}, fopt);
-    at::Tensor psdyx = at::empty({(int64_t)vg.row_blocks * C}, fopt);
-    if (bf16)
-      hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<__hip_bfloat16, 8>),
-                         dim3(vg.grid), dim3(256), 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         (const __hip_bfloat16*)dy.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         psdy.data_ptr<float>(), psdyx.data_ptr<float>(), M,
-                         vg.nC8, vg.c8_per_block, vg.row_blocks);
-    else
-      hipLaunchKernelGGL((bn_bwd_stats_vec_kernel<float, 4>), dim3(vg.grid),
-                         dim3(256), 0, stream, x.data_ptr<float>(),
-                         dy.data_ptr<float>(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), psdy.data_ptr<float>(),
-                         psdyx.data_ptr<float>(), M, vg.nC8, vg.c8_per_block,
-                         vg.row_blocks);
-    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(256),
-                       0, stream, psdy.data_ptr<float>(), psdyx.data_ptr<float>(),
-                       vg.row_blocks, sdy.data_ptr<float>(),
-                       sdyx.data_ptr<float>(), C);
-  } else {
-    sdy.zero_();
-    sdyx.zero_();
-    int c_blocks = (C + 63) / 64;
-    int row_blocks = std::max(1, 512 / c_blocks);
-    dim3 grid(c_blocks * row_blocks), block(256);
-    if (bf16)
-      hipLaunchKernelGGL(bn_bwd_stats_scalar_kernel<__hip_bfloat16>, grid, block,
-                         0, stream, (const __hip_bfloat16*)x.data_ptr(),
-                         (const __hip_bfloat16*)dy.data_ptr(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), M, C,
-                         c_blocks, row_blocks);
-    else
-      hipLaunchKernelGGL(bn_bwd_stats_scalar_kernel<float>, grid, block, 0,
-                         stream, x.data_ptr<float>(), dy.data_ptr<float>(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), M, C,
-                         c_blocks, row_blocks);
-  }
-
   at::Tensor dx = at::empty_like(x);
+  at::Tensor dres;
+  if (need_dres) dres = at::empty_like(x);
   float invM = 1.f / (float)M;
-  if (vg.ok) {
-    if (bf16)
-      hipLaunchKernelGGL((bn_bwd_apply_vec<__hip_bfloat16, 8>), dim3(vg.grid),
-                         dim3(256), 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         (const __hip_bfloat16*)dy.data_ptr(),
-                         (__hip_bfloat16*)dx.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM, M,
-                         vg.nC8, vg.c8_per_block, vg.row_blocks);
-    else
-      hipLaunchKernelGGL((bn_bwd_apply_vec<float, 4>), dim3(vg.grid), dim3(256),
-                         0, stream, x.data_ptr<float>(), dy.data_ptr<float>(),
-                         dx.data_ptr<float>(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM, M,
-                         vg.nC8, vg.c8_per_block, vg.row_blocks);
-  } else {
-    size_t total = (size_t)x.numel();
-    if (bf16)
-      hipLaunchKernelGGL(bn_bwd_apply_scalar<__hip_bfloat16>,
-                         dim3(scalar_grid(total)), dim3(256), 0, stream,
-                         (const __hip_bfloat16*)x.data_ptr(),
-                         (const __hip_bfloat16*)dy.data_ptr(),
-                         (__hip_bfloat16*)dx.data_ptr(), mean.data_ptr<float>(),
-                         invstd.data_ptr<float>(), weight.data_ptr<float>(),
-                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), invM,
-                         total, C);
-    else
-      hipLaunchKernelGGL(bn_bwd_apply_scalar<float>, dim3(scalar_grid(total)),
-                         dim3(256), 0, stream, x.data_ptr<float>(),
-                         dy.data_ptr<float>(), dx.data_ptr<float>(),
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         weight.data_ptr<float>(), sdy.data_ptr<float>(),
-                         sdyx.data_ptr<float>(), invM, total, C);
-  }
-  return {dx, sdyx, sdy};  // dweight = sum(dy*xhat), dbias = sum(dy)
+
+#define TMA_BN_BWD(T, VEC)                                                     \
+  do {                                                                         \
+    const T* xp = (const T*)x.data_ptr();                                      \
+    const T* gp = (const T*)dy.data_ptr();                                     \
+    const T* yp = relu ? (const T*)yout.data_ptr() : nullptr;                  \
+    T* dxp = (T*)dx.data_ptr();                                                \
+    T* drp = need_dres ? (T*)dres.data_ptr() : nullptr;                        \
+    if (vg.ok) {                                                               \
+      at::Tensor psdy = at::empty({(int64_t)vg.row_blocks * C}, fopt);         \
+      at::Tensor psdyx = at::empty({(int64_t)vg.row_blocks * C}, fopt);        \
+      auto sargs = [&](auto kern) {                                            \
+        hipLaunchKernelGGL(kern, dim3(vg.grid), dim3(256), 0, stream, xp, gp,  \
+                           yp, mean.data_ptr<float>(),                         \
+                           invstd.data_ptr<float>(), psdy.data_ptr<float>(),   \
+                           psdyx.data_ptr<float>(), M, vg.nC8,                 \
+                           vg.c8_per_block, vg.row_blocks);                    \
+      };                                                                       \
+      if (relu) sargs(bn_bwd_stats_vec_kernel<T, VEC, true>);                  \
+      else sargs(bn_bwd_stats_vec_kernel<T, VEC, false>);                      \
+      hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(C), dim3(256), 0,        \
+                         stream, psdy.data_ptr<float>(),                       \
+                         psdyx.data_ptr<float>(), vg.row_blocks,               \
+                         sdy.data_ptr<float>(), sdyx.data_ptr<float>(), C);    \
+      auto aargs = [&](auto kern) {                                            \
+        hipLaunchKernelGGL(kern, dim3(vg.grid), dim3(256), 0, stream, xp, gp,  \
+                           yp, dxp, drp, mean.data_ptr<float>(),               \
+                           invstd.data_ptr<float>(), weight.data_ptr<float>(), \
+                           sdy.data_ptr<float>(), sdyx.data_ptr<float>(),      \
+                           invM, M, vg.nC8, vg.c8_per_block, vg.row_blocks);   \
+      };                                                                       \
+      if (relu && need_dres) aargs(bn_bwd_apply_vec<T, VEC, true, true>);      \
+      else if (relu) aargs(bn_bwd_apply_vec<T, VEC, true, false>);             \
+      else if (need_dres) aargs(bn_bwd_apply_vec<T, VEC, false, true>);        \
+      else aargs(bn_bwd_apply_vec<T, VEC, false, false>);                      \
+    } else {                                                                   \
+      sdy.zero_();                                                             \
+      sdyx.zero_();                                                            \
+      int c_blocks = (C + 63) / 64;                                            \
+      int row_blocks = std::max(1, 512 / c_blocks);                            \
+      dim3 grid(c_blocks * row_blocks), block(256);                            \
+      auto sargs = [&](auto kern) {                                            \
+        hipLaunchKernelGGL(kern, grid, block, 0, stream, xp, gp, yp,           \
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),   \
+                           sdy.data_ptr<float>(), sdyx.data_ptr<float>(), M,   \
+                           C, c_blocks, row_blocks);                           \
+      };                                                                       \
+      if (relu) sargs(bn_bwd_stats_scalar_kernel<T, true>);                    \
+      else sargs(bn_bwd_stats_scalar_kernel<T, false>);                        \
+      size_t total = (size_t)x.numel();                                        \
+      auto aargs = [&](auto kern) {                                            \
+        hipLaunchKernelGGL(kern, dim3(scalar_grid(total)), dim3(256), 0,       \
+                           stream, xp, gp, yp, dxp, drp,                       \
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),   \
+                           weight.data_ptr<float>(), sdy.data_ptr<float>(),    \
+                           sdyx.data_ptr<float>(), invM, total, C);            \
+      };                                                                       \
+      if (relu && need_dres) aargs(bn_bwd_apply_scalar<T, true, true>);        \
+      else if (relu) aargs(bn_bwd_apply_scalar<T, true, false>);               \
+      else if (need_dres) aargs(bn_bwd_apply_scalar<T, false, true>);          \
+      else aargs(bn_bwd_apply_scalar<T, false, false>);                        \
+    }                                                                          \
+  } while (0)
+  if (bf16)
+    TMA_BN_BWD(__hip_bfloat16, 8);
+  else
+    TMA_BN_BWD(float, 4);
+#undef TMA_BN_BWD
+  if (need_dres)
+    return {dx, sdyx, sdy, dres};  // dweight = sum(g*xhat), dbias = sum(g)
+  return {dx, sdyx, sdy};
 }
 
 void register_batchnorm(pybind11::module_& m) {
-  m.def("bn_forward", &bn_forward);
-  m.def("bn_backward", &bn_backward);
+  namespace py = pybind11;
+  m.def("bn_forward", &bn_forward, py::arg("x"), py::arg("weight"),
+        py::arg("bias"), py::arg("running_mean"), py::arg("running_var"),
+        py::arg("training"), py::arg("momentum"), py::arg("eps"),
+        py::arg("relu") = false, py::arg("res") = py::none());
+  m.def("bn_backward", &bn_backward, py::arg("x"), py::arg("dy"),
+        py::arg("weight"), py::arg("mean"), py::arg("invstd"),
+        py::arg("yout") = py::none(), py::arg("relu") = false,
+        py::arg("need_dres") = false);
 }
 
 }  // namespace tma

@@ -24,6 +24,14 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x
+        if getattr(self.bn1, "supports_fusion", False) and x.is_cuda:
+            # FusedBatchNorm2d path: relu and the residual add run inside
+            # the BN epilogue kernels (no separate elementwise passes).
+            out = self.bn1(self.conv1(x), relu=True)
+            out = self.bn2(self.conv2(out), relu=True)
+            if self.downsample is not None:
+                identity = self.downsample(x)
+            return self.bn3(self.conv3(out), relu=True, res=identity)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))
         out = self.bn3(self.conv3(out))
@@ -69,7 +77,10 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        if getattr(self.bn1, "supports_fusion", False) and x.is_cuda:
+            x = self.maxpool(self.bn1(self.conv1(x), relu=True))
+        else:
+            x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         x = self.avgpool(x).flatten(1)
         return self.fc(x)

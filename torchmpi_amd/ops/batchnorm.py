@@ -3,6 +3,12 @@
 channels_last bf16/f32 tensors — MIOpen's spatial BN is ~35% of the
 ResNet-50 step on MI355X (profiles/).
 
+The kernels optionally fuse the FOLLOWING relu and a residual add into the
+BN epilogue (y = relu(bn(x) + res)): the separate elementwise add/relu
+passes were ~20% of the ResNet-50 step (profiles/
+resnet50_bf16_kernel_stats.md). Modules that know their successor call
+`forward(x, relu=True, res=identity)` (see models/resnet.py Bottleneck).
+
 State-dict compatible with nn.BatchNorm2d; convert_batchnorm(model) swaps
 all BatchNorm2d modules in place.
 """
@@ -16,37 +22,53 @@ from .. import _C
 
 class _BnFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, running_mean, running_var, training,
-                momentum, eps):
+    def forward(ctx, x, res, weight, bias, running_mean, running_var, training,
+                momentum, eps, relu):
         x = x.contiguous(memory_format=torch.channels_last)
+        if res is not None:
+            res = res.contiguous(memory_format=torch.channels_last)
         y, mean, invstd = _C.bn_forward(x, weight, bias, running_mean,
-                                        running_var, training, momentum, eps)
-        ctx.save_for_backward(x, weight, mean, invstd)
+                                        running_var, training, momentum, eps,
+                                        relu, res)
+        # y is only needed for backward when the relu mask must be recovered.
+        ctx.save_for_backward(x, weight, mean, invstd,
+                              y if relu else torch.empty(0))
         ctx.training = training
+        ctx.relu = relu
+        ctx.has_res = res is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, mean, invstd = ctx.saved_tensors
+        x, weight, mean, invstd, y = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
         if not ctx.training:
             # eval: stats are constants — no batch-statistic terms.
+            dyf = dy.float()
+            if ctx.relu:
+                dyf = dyf * (y.float() > 0)
             m = mean.view(1, -1, 1, 1)
             istd = invstd.view(1, -1, 1, 1)
             xhat = (x.float() - m) * istd
-            dyf = dy.float()
             dx = (dyf * (weight.view(1, -1, 1, 1) * istd)).to(x.dtype)
             dweight = (dyf * xhat).sum((0, 2, 3))
             dbias = dyf.sum((0, 2, 3))
-            return (dx.contiguous(memory_format=torch.channels_last), dweight,
-                    dbias, None, None, None, None, None)
-        dx, dweight, dbias = _C.bn_backward(x, dy, weight, mean, invstd)
-        return dx, dweight, dbias, None, None, None, None, None
+            dres = dyf.to(x.dtype) if ctx.has_res else None
+            return (dx.contiguous(memory_format=torch.channels_last), dres,
+                    dweight, dbias, None, None, None, None, None, None)
+        out = _C.bn_backward(x, dy, weight, mean, invstd,
+                             y if ctx.relu else None, ctx.relu, ctx.has_res)
+        dres = out[3] if ctx.has_res else None
+        return (out[0], dres, out[1], out[2], None, None, None, None, None,
+                None)
 
 
 class FusedBatchNorm2d(nn.Module):
     """Per-channel batchnorm with fp32 stats/affine and bf16/f32 NHWC data
-    path on our kernels; falls back to torch BN on CPU / non-4D input."""
+    path on our kernels; falls back to torch BN on CPU / non-4D input.
+    forward(x, relu=..., res=...) fuses a following ReLU / residual add."""
+
+    supports_fusion = True  # duck-typed by models that fuse relu/residual
 
     def __init__(self, num_features: int, eps: float = 1e-5,
                  momentum: float = 0.1):
@@ -61,11 +83,15 @@ class FusedBatchNorm2d(nn.Module):
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
 
-    def forward(self, x):
+    def forward(self, x, relu: bool = False,
+                res: Optional[torch.Tensor] = None):
         if not x.is_cuda or x.dim() != 4:
-            return torch.nn.functional.batch_norm(
+            y = torch.nn.functional.batch_norm(
                 x, self.running_mean, self.running_var, self.weight, self.bias,
                 self.training, self.momentum, self.eps)
+            if res is not None:
+                y = y + res
+            return torch.relu(y) if relu else y
         if self.training:
             self.num_batches_tracked += 1
         # autocast: keep our fp32-stat kernels; input stays in its dtype
@@ -73,9 +99,13 @@ class FusedBatchNorm2d(nn.Module):
             xw = x
             if xw.dtype not in (torch.bfloat16, torch.float32):
                 xw = xw.to(torch.bfloat16)
-            return _BnFunction.apply(xw, self.weight, self.bias,
+            rw = res
+            if rw is not None and rw.dtype != xw.dtype:
+                rw = rw.to(xw.dtype)
+            return _BnFunction.apply(xw, rw, self.weight, self.bias,
                                      self.running_mean, self.running_var,
-                                     self.training, self.momentum, self.eps)
+                                     self.training, self.momentum, self.eps,
+                                     relu)
 
     @classmethod
     def from_batchnorm(cls, bn: nn.BatchNorm2d) -> "FusedBatchNorm2d":
