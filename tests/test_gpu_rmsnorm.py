@@ -61,3 +61,72 @@ def test_rmsnorm_in_llama_block():
     torch.cuda.synchronize()
     assert torch.isfinite(loss)
     assert all(p.grad is not None for p in model.parameters())
+
+
+@requires_gpu
+@pytest.mark.parametrize("shape", [(4, 128, 4096), (2, 7, 256)])
+def test_fused_add_rmsnorm(shape):
+    """(y, h) = (rmsnorm(res+branch), res+branch) fused vs eager fp32,
+    forward + backward through BOTH outputs (h feeds downstream)."""
+    from torchmpi_amd.ops.rmsnorm import fused_add_rmsnorm
+    torch.manual_seed(1)
+    dev = torch.device("cuda", 0)
+    eps = 1e-5
+    res = torch.randn(shape, device=dev, dtype=torch.bfloat16)
+    br = torch.randn(shape, device=dev, dtype=torch.bfloat16)
+    w = (torch.randn(shape[-1], device=dev) * 0.1 + 1.0)
+
+    r_f = res.clone().requires_grad_(True)
+    b_f = br.clone().requires_grad_(True)
+    w_f = w.clone().requires_grad_(True)
+    r_r = res.clone().float().requires_grad_(True)
+    b_r = br.clone().float().requires_grad_(True)
+    w_r = w.clone().requires_grad_(True)
+
+    y_f, h_f = fused_add_rmsnorm(r_f, b_f, w_f, eps)
+    h_r = r_r + b_r
+    y_r = eager_rms(h_r, w_r, eps)
+    torch.cuda.synchronize()
+    assert torch.allclose(h_f.float(), h_r, atol=2e-2, rtol=2e-2)
+    assert torch.allclose(y_f.float(), y_r, atol=5e-2, rtol=5e-2), \
+        (y_f.float() - y_r).abs().max()
+
+    gy = torch.randn_like(y_r)
+    gh = torch.randn_like(h_r)
+    # downstream uses both outputs (like the next block's attn + residual)
+    (y_f.float() * gy + h_f.float() * gh).sum().backward()
+    (y_r * gy + h_r * gh).sum().backward()
+    torch.cuda.synchronize()
+    assert torch.allclose(r_f.grad.float(), r_r.grad, atol=1e-1, rtol=5e-2), \
+        (r_f.grad.float() - r_r.grad).abs().max()
+    assert torch.allclose(b_f.grad.float(), b_r.grad, atol=1e-1, rtol=5e-2)
+    assert torch.allclose(w_f.grad, w_r.grad, atol=5e-1, rtol=5e-2), \
+        (w_f.grad - w_r.grad).abs().max()
+
+
+@requires_gpu
+def test_llama_block_pair_equivalence():
+    """The (res, branch)-pair Block forward == the textbook
+    x + attn(ln1(x)); x + mlp(ln2(x)) composition."""
+    from torchmpi_amd.models.llama import Block, _add_norm, llama_debug
+    torch.manual_seed(2)
+    dev = torch.device("cuda", 0)
+    cfg = llama_debug()
+    blk = Block(cfg).to(dev).to(torch.bfloat16)
+    for p in blk.parameters():
+        if p.dim() == 1:
+            p.data = p.data.float()  # norm weights fp32 (fused path)
+    blk.ln1.weight.data = blk.ln1.weight.data.float()
+    blk.ln2.weight.data = blk.ln2.weight.data.float()
+    from torchmpi_amd.models.llama import _rope_freqs
+    freqs = _rope_freqs(cfg.dim // cfg.n_heads, 64, cfg.rope_theta, dev)
+    x = torch.randn(2, 16, cfg.dim, device=dev, dtype=torch.bfloat16)
+    with torch.no_grad():
+        res, branch = blk(x, None, freqs)
+        fused_out = (res + branch).float()
+        # reference composition with the same weights
+        ref = x + blk.attn(blk.ln1(x), freqs)
+        ref = (ref + blk.mlp(blk.ln2(ref))).float()
+    torch.cuda.synchronize()
+    assert torch.allclose(fused_out, ref, atol=1e-1, rtol=5e-2), \
+        (fused_out - ref).abs().max()

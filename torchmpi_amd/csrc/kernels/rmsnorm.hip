@@ -52,12 +52,19 @@ __device__ __forceinline__ float block_sum(float v, float* lds) {
 
 // ---- forward: one block per row; x stays in registers ---------------------
 // D must be a multiple of 8 and <= 256 * PERT * 8. PERT = elems/8 per thread.
-template <int PERT>
+// ADD fuses the Llama pre-norm residual add: h = x + res (bf16-rounded,
+// written to hout and used for the stats/scale so backward sees exactly the
+// forward values), y = rmsnorm(h) — saves the separate add pass and the
+// norm's re-read of the sum.
+template <int PERT, bool ADD>
 __global__ __launch_bounds__(256) void rms_fwd_kernel(
-    const __hip_bfloat16* __restrict__ x, __hip_bfloat16* __restrict__ y,
+    const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ res,
+    __hip_bfloat16* __restrict__ hout, __hip_bfloat16* __restrict__ y,
     const float* __restrict__ w, float* __restrict__ inv_out, int D, float eps) {
   const size_t row = blockIdx.x;
   const __hip_bfloat16* xr = x + row * (size_t)D;
+  const __hip_bfloat16* rr = ADD ? res + row * (size_t)D : nullptr;
+  __hip_bfloat16* hr = ADD ? hout + row * (size_t)D : nullptr;
   __hip_bfloat16* yr = y + row * (size_t)D;
   __hip_bfloat16 v[PERT][8];
   float ss = 0.f;
@@ -67,6 +74,14 @@ __global__ __launch_bounds__(256) void rms_fwd_kernel(
     int i = (int)threadIdx.x + k * 256;
     if (i < nvec) {
       *reinterpret_cast<uint4*>(v[k]) = reinterpret_cast<const uint4*>(xr)[i];
+      if (ADD) {
+        __hip_bfloat16 rv[8];
+        *reinterpret_cast<uint4*>(rv) = reinterpret_cast<const uint4*>(rr)[i];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[k][e] = __float2bfloat16(bf2f(v[k][e]) + bf2f(rv[e]));
+        reinterpret_cast<uint4*>(hr)[i] = *reinterpret_cast<uint4*>(v[k]);
+      }
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         float f = bf2f(v[k][e]);
@@ -93,14 +108,17 @@ __global__ __launch_bounds__(256) void rms_fwd_kernel(
 }
 
 // ---- backward dx: one block per row ---------------------------------------
-template <int PERT>
+// ADD: dres (the downstream gradient of the residual sum h) is added into
+// dx in the same pass — the fused op's both inputs receive this tensor.
+template <int PERT, bool ADD>
 __global__ __launch_bounds__(256) void rms_bwd_dx_kernel(
     const __hip_bfloat16* __restrict__ x, const __hip_bfloat16* __restrict__ dy,
-    __hip_bfloat16* __restrict__ dx, const float* __restrict__ w,
-    const float* __restrict__ inv_in, int D) {
+    const __hip_bfloat16* __restrict__ dres, __hip_bfloat16* __restrict__ dx,
+    const float* __restrict__ w, const float* __restrict__ inv_in, int D) {
   const size_t row = blockIdx.x;
   const __hip_bfloat16* xr = x + row * (size_t)D;
   const __hip_bfloat16* gr = dy + row * (size_t)D;
+  const __hip_bfloat16* er = ADD ? dres + row * (size_t)D : nullptr;
   __hip_bfloat16* or_ = dx + row * (size_t)D;
   const float inv = inv_in[row];
   __hip_bfloat16 xv[PERT][8], gv[PERT][8];
@@ -125,12 +143,16 @@ __global__ __launch_bounds__(256) void rms_bwd_dx_kernel(
   for (int k = 0; k < PERT; ++k) {
     int i = (int)threadIdx.x + k * 256;
     if (i < nvec) {
-      __hip_bfloat16 o[8];
+      __hip_bfloat16 o[8], ev[8];
+      if (ADD)
+        *reinterpret_cast<uint4*>(ev) = reinterpret_cast<const uint4*>(er)[i];
       const float* wd = w + i * 8;
 #pragma unroll
-      for (int e = 0; e < 8; ++e)
-        o[e] = __float2bfloat16(inv * wd[e] * bf2f(gv[k][e]) -
-                                bf2f(xv[k][e]) * c);
+      for (int e = 0; e < 8; ++e) {
+        float g = inv * wd[e] * bf2f(gv[k][e]) - bf2f(xv[k][e]) * c;
+        if (ADD) g += bf2f(ev[e]);
+        o[e] = __float2bfloat16(g);
+      }
       reinterpret_cast<uint4*>(or_)[i] = *reinterpret_cast<uint4*>(o);
     }
   }
@@ -222,10 +244,11 @@ std::vector<at::Tensor> rms_forward(at::Tensor x, at::Tensor weight, double eps)
   const int nvec = D / 8;
   const int pert = (nvec + 255) / 256;
 #define TMA_RMS_FWD(P)                                                        \
-  hipLaunchKernelGGL((rms_fwd_kernel<P>), dim3((uint32_t)R), dim3(256), 0,    \
-                     stream, (const __hip_bfloat16*)x.data_ptr(),             \
-                     (__hip_bfloat16*)y.data_ptr(), weight.data_ptr<float>(), \
-                     inv.data_ptr<float>(), D, (float)eps)
+  hipLaunchKernelGGL((rms_fwd_kernel<P, false>), dim3((uint32_t)R), dim3(256), \
+                     0, stream, (const __hip_bfloat16*)x.data_ptr(), nullptr,  \
+                     nullptr, (__hip_bfloat16*)y.data_ptr(),                   \
+                     weight.data_ptr<float>(), inv.data_ptr<float>(), D,       \
+                     (float)eps)
   switch (pert) {
     case 1: TMA_RMS_FWD(1); break;
     case 2: TMA_RMS_FWD(2); break;
@@ -236,28 +259,78 @@ std::vector<at::Tensor> rms_forward(at::Tensor x, at::Tensor weight, double eps)
   return {y, inv};
 }
 
-// Returns (dx, dweight[fp32]).
+// Fused h = x + res; y = rmsnorm(h). Returns (y, h, inv) — h is the new
+// residual stream (bf16-rounded exactly as used for the stats).
+std::vector<at::Tensor> rms_forward_add(at::Tensor x, at::Tensor res,
+                                        at::Tensor weight, double eps) {
+  const int D = (int)x.size(-1);
+  TMA_CHECK(rms_supported(x, D), "rms_forward_add: unsupported input");
+  TMA_CHECK(res.sizes() == x.sizes() && res.scalar_type() == x.scalar_type() &&
+                res.is_contiguous(),
+            "rms_forward_add: residual mismatch");
+  TMA_CHECK(weight.scalar_type() == at::kFloat && weight.is_contiguous(),
+            "rms_forward_add: fp32 contiguous weight");
+  const size_t R = (size_t)x.numel() / D;
+  auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  at::Tensor y = at::empty_like(x);
+  at::Tensor h = at::empty_like(x);
+  at::Tensor inv = at::empty({(int64_t)R},
+                             at::TensorOptions().dtype(at::kFloat).device(x.device()));
+  const int nvec = D / 8;
+  const int pert = (nvec + 255) / 256;
+#define TMA_RMS_FWDA(P)                                                        \
+  hipLaunchKernelGGL((rms_fwd_kernel<P, true>), dim3((uint32_t)R), dim3(256),  \
+                     0, stream, (const __hip_bfloat16*)x.data_ptr(),           \
+                     (const __hip_bfloat16*)res.data_ptr(),                    \
+                     (__hip_bfloat16*)h.data_ptr(),                            \
+                     (__hip_bfloat16*)y.data_ptr(), weight.data_ptr<float>(),  \
+                     inv.data_ptr<float>(), D, (float)eps)
+  switch (pert) {
+    case 1: TMA_RMS_FWDA(1); break;
+    case 2: TMA_RMS_FWDA(2); break;
+    case 3: TMA_RMS_FWDA(3); break;
+    default: TMA_RMS_FWDA(4); break;
+  }
+#undef TMA_RMS_FWDA
+  return {y, h, inv};
+}
+
+// Returns (dx, dweight[fp32]). `dres` (optional) is the downstream gradient
+// of the residual sum h — fused into dx so both inputs of the fused add+norm
+// receive one tensor.
 std::vector<at::Tensor> rms_backward(at::Tensor x, at::Tensor dy, at::Tensor weight,
-                                     at::Tensor inv) {
+                                     at::Tensor inv,
+                                     c10::optional<at::Tensor> dres_opt) {
   const int D = (int)x.size(-1);
   TMA_CHECK(rms_supported(x, D) && dy.is_contiguous(),
             "rms_backward: unsupported input");
+  at::Tensor dres = dres_opt.value_or(at::Tensor());
+  const bool add = dres.defined();
+  if (add)
+    TMA_CHECK(dres.sizes() == x.sizes() && dres.is_contiguous() &&
+                  dres.scalar_type() == x.scalar_type(),
+              "rms_backward: dres mismatch");
   const size_t R = (size_t)x.numel() / D;
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   at::Tensor dx = at::empty_like(x);
   const int nvec = D / 8;
   const int pert = (nvec + 255) / 256;
-#define TMA_RMS_DX(P)                                                          \
-  hipLaunchKernelGGL((rms_bwd_dx_kernel<P>), dim3((uint32_t)R), dim3(256), 0,  \
-                     stream, (const __hip_bfloat16*)x.data_ptr(),              \
+#define TMA_RMS_DX(P, A)                                                       \
+  hipLaunchKernelGGL((rms_bwd_dx_kernel<P, A>), dim3((uint32_t)R), dim3(256),  \
+                     0, stream, (const __hip_bfloat16*)x.data_ptr(),           \
                      (const __hip_bfloat16*)dy.data_ptr(),                     \
+                     add ? (const __hip_bfloat16*)dres.data_ptr() : nullptr,   \
                      (__hip_bfloat16*)dx.data_ptr(), weight.data_ptr<float>(), \
                      inv.data_ptr<float>(), D)
-  switch (pert) {
-    case 1: TMA_RMS_DX(1); break;
-    case 2: TMA_RMS_DX(2); break;
-    case 3: TMA_RMS_DX(3); break;
-    default: TMA_RMS_DX(4); break;
+  switch (pert * 2 + (add ? 1 : 0)) {
+    case 2: TMA_RMS_DX(1, false); break;
+    case 3: TMA_RMS_DX(1, true); break;
+    case 4: TMA_RMS_DX(2, false); break;
+    case 5: TMA_RMS_DX(2, true); break;
+    case 6: TMA_RMS_DX(3, false); break;
+    case 7: TMA_RMS_DX(3, true); break;
+    case 8: TMA_RMS_DX(4, false); break;
+    default: TMA_RMS_DX(4, true); break;
   }
 #undef TMA_RMS_DX
 
@@ -295,9 +368,12 @@ std::vector<at::Tensor> rms_backward(at::Tensor x, at::Tensor dy, at::Tensor wei
 }
 
 void register_rmsnorm(pybind11::module_& m) {
+  namespace py = pybind11;
   m.def("rms_ok", &rms_ok);
   m.def("rms_forward", &rms_forward);
-  m.def("rms_backward", &rms_backward);
+  m.def("rms_forward_add", &rms_forward_add);
+  m.def("rms_backward", &rms_backward, py::arg("x"), py::arg("dy"),
+        py::arg("weight"), py::arg("inv"), py::arg("dres") = py::none());
 }
 
 }  // namespace tma

@@ -53,6 +53,24 @@ class RMSNorm(nn.Module):
         return (x * self.weight.float()).to(dt)
 
 
+def _add_norm(norm: RMSNorm, res, branch):
+    """(normed, new_res) = (rmsnorm(res + branch), res + branch); the add is
+    fused into the norm kernel on bf16 CUDA (csrc/kernels/rmsnorm.hip) —
+    the residual stream never takes a separate elementwise pass. branch may
+    be None (start of the stream)."""
+    if branch is None:
+        return norm(res), res
+    if res.is_cuda:
+        from ..ops.rmsnorm import fused_add_rmsnorm, fused_rmsnorm_supported
+        rc = res.contiguous()
+        bc = branch.contiguous()
+        if (norm.weight.dtype == torch.float32 and bc.dtype == rc.dtype
+                and fused_rmsnorm_supported(rc)):
+            return fused_add_rmsnorm(rc, bc, norm.weight, norm.eps)
+    h = res + branch
+    return norm(h), h
+
+
 def _rope_freqs(dim, max_seq, theta, device):
     inv = 1.0 / (theta ** (torch.arange(0, dim, 2, device=device).float() / dim))
     t = torch.arange(max_seq, device=device).float()
@@ -111,9 +129,14 @@ class Block(nn.Module):
         self.ln1 = RMSNorm(cfg.dim, cfg.norm_eps)
         self.ln2 = RMSNorm(cfg.dim, cfg.norm_eps)
 
-    def forward(self, x, freqs):
-        x = x + self.attn(self.ln1(x), freqs)
-        return x + self.mlp(self.ln2(x))
+    def forward(self, res, branch, freqs):
+        """Carries the residual stream as a (res, pending-branch) pair so the
+        pre-norm adds fuse into the norm kernels: equivalent to
+        x = x + attn(ln1(x)); x = x + mlp(ln2(x))."""
+        n1, res = _add_norm(self.ln1, res, branch)
+        a = self.attn(n1, freqs)
+        n2, res = _add_norm(self.ln2, res, a)
+        return res, self.mlp(n2)
 
 
 class LlamaModel(nn.Module):
@@ -141,9 +164,10 @@ class LlamaModel(nn.Module):
             # bf16 residual stream (standard bf16 mixed precision; RMSNorm
             # stats still accumulate in fp32 inside the fused kernel).
             x = x.to(torch.bfloat16)
+        branch = None
         for blk in self.blocks:
-            x = blk(x, self.freqs)
-        x = self.norm(x)
+            x, branch = blk(x, branch, self.freqs)
+        x, _ = _add_norm(self.norm, x, branch)
         if targets is not None:
             logits = self.head(x)
             return F.cross_entropy(logits.float().view(-1, logits.size(-1)),
