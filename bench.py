@@ -15,6 +15,8 @@ Prints ONE JSON line from rank 0:
 import argparse
 import json
 import os
+import sys
+import threading
 import time
 
 # MIOpen algorithm search: FAST find gives the same steady-state kernels as
@@ -33,6 +35,36 @@ def parse_args():
     p.add_argument("--local-batch", type=int, default=256)
     p.add_argument("--bucket-mb", type=int, default=25)
     return p.parse_args()
+
+
+class _Watchdog:
+    """First-contact safety: if engine init / the first collectives hang
+    (e.g. a grouped-p2p mismatch on a topology never seen before), exit 124
+    with a diagnostic instead of wedging the whole job (round-1 verdict
+    item 2). Disarmed once steady state is reached."""
+
+    def __init__(self, seconds):
+        self.seconds = seconds
+        self.stage = None
+        self.deadline = None
+        threading.Thread(target=self._run, daemon=True).start()
+
+    def arm(self, stage):
+        self.stage = stage
+        self.deadline = time.monotonic() + self.seconds
+
+    def disarm(self):
+        self.deadline = None
+
+    def _run(self):
+        while True:
+            time.sleep(2.0)
+            if self.deadline is not None and time.monotonic() > self.deadline:
+                sys.stderr.write(
+                    f"[bench watchdog] rank {os.environ.get('RANK')} stage "
+                    f"'{self.stage}' exceeded {self.seconds}s — aborting\n")
+                sys.stderr.flush()
+                os._exit(124)
 
 
 def verify_allreduce(tm, device):
@@ -135,7 +167,10 @@ def main():
                                 comm_dtype=torch.bfloat16 if use_gpu else None)
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
 
+    wd = _Watchdog(float(os.environ.get("TORCHMPI_BENCH_STAGE_TIMEOUT", "600")))
+    wd.arm("engine init + allreduce verify (first 8-rank contact)")
     graph_status = verify_allreduce(tm, device)
+    wd.disarm()
 
     img = 224 if use_gpu else 64
     x = torch.randn(local_batch, 3, img, img, device=device, dtype=torch.float32)
@@ -153,7 +188,14 @@ def main():
         opt.step()
         return loss
 
-    for _ in range(args.warmup):
+    # First training step covers the bucketed-overlap path's first contact
+    # (hooks + grouped p2p); steady state afterwards runs un-watchdogged.
+    wd.arm("first training step (bucketed allreduce first contact)")
+    step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    wd.disarm()
+    for _ in range(max(args.warmup - 1, 0)):
         step()
     if use_gpu:
         torch.cuda.synchronize()
@@ -174,7 +216,9 @@ def main():
     all_el = tm.allgather(el_t)
     elapsed_max = float(all_el.max())
 
+    wd.arm("256MB busbw measurement")
     busbw = measure_allreduce_busbw(tm, device)
+    wd.disarm()
 
     if rank == 0:
         samples_per_sec = W * local_batch * args.steps / elapsed_max
