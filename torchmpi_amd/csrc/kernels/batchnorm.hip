@@ -257,9 +257,10 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
 template <typename T, int VEC, bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_fwd_apply_vec(
     const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
-    const float* __restrict__ mean, const float* __restrict__ invstd,
-    const float* __restrict__ weight, const float* __restrict__ bias, size_t M,
-    int nC8, int c8_per_block, int row_blocks) {
+    uint8_t* __restrict__ mask, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ weight,
+    const float* __restrict__ bias, size_t M, int nC8, int c8_per_block,
+    int row_blocks) {
   const int c_blocks = nC8 / c8_per_block;
   const int cb = blockIdx.x % c_blocks;
   const int rb = blockIdx.x / c_blocks;
@@ -286,14 +287,21 @@ __global__ __launch_bounds__(256) void bn_fwd_apply_vec(
     if (RES)
       *reinterpret_cast<uint4*>(rv) =
           *reinterpret_cast<const uint4*>(rb2 + r * (size_t)C);
+    unsigned mb = 0;
 #pragma unroll
     for (int e = 0; e < VEC; ++e) {
       float f = ldf(v[e]) * sc[e] + sh[e];
       if (RES) f += ldf(rv[e]);
-      if (RELU) f = fmaxf(f, 0.f);
+      if (RELU) {
+        mb |= (unsigned)(f > 0.f) << e;
+        f = fmaxf(f, 0.f);
+      }
       o[e] = stf<T>(f);
     }
     *reinterpret_cast<uint4*>(yb + r * (size_t)C) = *reinterpret_cast<uint4*>(o);
+    // 1-bit relu mask, one byte per VEC elements: backward reads this
+    // instead of re-streaming y (saves ~2 activation passes per step).
+    if (RELU) mask[r * (size_t)nC8 + c8] = (uint8_t)mb;
   }
 }
 
@@ -320,7 +328,7 @@ __global__ __launch_bounds__(256) void bn_fwd_apply_scalar(
 template <typename T, int VEC, bool RELU>
 __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
     const T* __restrict__ x, const T* __restrict__ dy,
-    const T* __restrict__ yout, const float* __restrict__ mean,
+    const uint8_t* __restrict__ mask, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ psdy,
     float* __restrict__ psdyx, size_t M, int nC8, int c8_per_block,
     int row_blocks) {
@@ -342,20 +350,17 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_vec_kernel(
   const size_t row_stride = (size_t)row_blocks * rows_per_iter;
   const T* xb = x + (size_t)c8 * VEC;
   const T* gb = dy + (size_t)c8 * VEC;
-  const T* yb = RELU ? yout + (size_t)c8 * VEC : nullptr;
   for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
-    T xv[VEC], gv[VEC], yv[VEC];
+    T xv[VEC], gv[VEC];
     *reinterpret_cast<uint4*>(xv) =
         *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
     *reinterpret_cast<uint4*>(gv) =
         *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
-    if (RELU)
-      *reinterpret_cast<uint4*>(yv) =
-          *reinterpret_cast<const uint4*>(yb + r * (size_t)C);
+    unsigned mb = RELU ? mask[r * (size_t)nC8 + c8] : 0u;
 #pragma unroll
     for (int e = 0; e < VEC; ++e) {
       float g = ldf(gv[e]);
-      if (RELU && ldf(yv[e]) <= 0.f) g = 0.f;  // relu mask from fused output
+      if (RELU && !((mb >> e) & 1u)) g = 0.f;  // relu mask from forward
       float xh = (ldf(xv[e]) - mu[e]) * is[e];
       acc[e] += g;
       accx[e] += g * xh;
@@ -439,7 +444,7 @@ __global__ __launch_bounds__(256) void bn_bwd_stats_scalar_kernel(
 template <typename T, int VEC, bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_bwd_apply_vec(
     const T* __restrict__ x, const T* __restrict__ dy,
-    const T* __restrict__ yout, T* __restrict__ dx, T* __restrict__ dres,
+    const uint8_t* __restrict__ mask, T* __restrict__ dx, T* __restrict__ dres,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ sdy,
     const float* __restrict__ sdyx, float invM, size_t M, int nC8,
@@ -466,22 +471,19 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_vec(
   const size_t row_stride = (size_t)row_blocks * rows_per_iter;
   const T* xb = x + (size_t)c8 * VEC;
   const T* gb = dy + (size_t)c8 * VEC;
-  const T* yb = RELU ? yout + (size_t)c8 * VEC : nullptr;
   T* ob = dx + (size_t)c8 * VEC;
   T* eb = RES ? dres + (size_t)c8 * VEC : nullptr;
   for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
-    T xv[VEC], gv[VEC], yv[VEC], o[VEC], ev[VEC];
+    T xv[VEC], gv[VEC], o[VEC], ev[VEC];
     *reinterpret_cast<uint4*>(xv) =
         *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
     *reinterpret_cast<uint4*>(gv) =
         *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
-    if (RELU)
-      *reinterpret_cast<uint4*>(yv) =
-          *reinterpret_cast<const uint4*>(yb + r * (size_t)C);
+    unsigned mb = RELU ? mask[r * (size_t)nC8 + c8] : 0u;
 #pragma unroll
     for (int e = 0; e < VEC; ++e) {
       float g = ldf(gv[e]);
-      if (RELU && ldf(yv[e]) <= 0.f) g = 0.f;
+      if (RELU && !((mb >> e) & 1u)) g = 0.f;
       o[e] = stf<T>(A[e] * g + B[e] * ldf(xv[e]) + D[e]);
       if (RES) ev[e] = stf<T>(g);
     }
@@ -607,15 +609,22 @@ std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor b
   }
 
   at::Tensor y = at::empty_like(x);
+  // 1-bit relu mask (one byte per VEC elements) for the backward kernels —
+  // replaces re-streaming y; only the vector path uses it.
+  at::Tensor mask;
+  if (relu && vg.ok)
+    mask = at::empty({(int64_t)(M * (size_t)vg.nC8)},
+                     at::TensorOptions().dtype(at::kByte).device(x.device()));
 #define TMA_BN_FWD(T, VEC)                                                     \
   do {                                                                         \
     const T* xp = (const T*)x.data_ptr();                                      \
     const T* rp = has_res ? (const T*)res.data_ptr() : nullptr;                \
     T* yp = (T*)y.data_ptr();                                                  \
+    uint8_t* mp = mask.defined() ? mask.data_ptr<uint8_t>() : nullptr;         \
     if (vg.ok) {                                                               \
       auto args = [&](auto kern) {                                             \
         hipLaunchKernelGGL(kern, dim3(vg.grid), dim3(256), 0, stream, xp, rp,  \
-                           yp, mean.data_ptr<float>(),                         \
+                           yp, mp, mean.data_ptr<float>(),                     \
                            invstd.data_ptr<float>(), weight.data_ptr<float>(), \
                            bias.data_ptr<float>(), M, vg.nC8, vg.c8_per_block, \
                            vg.row_blocks);                                     \
@@ -643,7 +652,9 @@ std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor b
   else
     TMA_BN_FWD(float, 4);
 #undef TMA_BN_FWD
-  return {y, mean, invstd};
+  if (!mask.defined())
+    mask = at::empty({0}, at::TensorOptions().dtype(at::kByte).device(x.device()));
+  return {y, mean, invstd, mask};
 }
 
 // Returns (dx, dweight, dbias[, dres]). `yout` is the fused forward output
@@ -652,14 +663,13 @@ std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor b
 std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weight,
                                     at::Tensor mean, at::Tensor invstd,
                                     c10::optional<at::Tensor> yout_opt, bool relu,
-                                    bool need_dres) {
+                                    bool need_dres,
+                                    c10::optional<at::Tensor> mask_opt) {
   at::Tensor yout = yout_opt.value_or(at::Tensor());
+  at::Tensor mask = mask_opt.value_or(at::Tensor());
+  if (mask.defined() && mask.numel() == 0) mask = at::Tensor();
   check_bn_input(x);
   check_bn_input(dy);
-  if (relu) {
-    TMA_CHECK(yout.defined(), "bn_backward: relu needs the forward output");
-    check_bn_input(yout);
-  }
   auto stream = c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   const int C = (int)x.size(1);
   const size_t M = (size_t)x.numel() / C;
@@ -678,15 +688,19 @@ std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weig
   do {                                                                         \
     const T* xp = (const T*)x.data_ptr();                                      \
     const T* gp = (const T*)dy.data_ptr();                                     \
-    const T* yp = relu ? (const T*)yout.data_ptr() : nullptr;                  \
+    const T* yp = (relu && yout.defined()) ? (const T*)yout.data_ptr()         \
+                                           : nullptr;                          \
+    const uint8_t* mp =                                                        \
+        (relu && mask.defined()) ? mask.data_ptr<uint8_t>() : nullptr;         \
     T* dxp = (T*)dx.data_ptr();                                                \
     T* drp = need_dres ? (T*)dres.data_ptr() : nullptr;                        \
     if (vg.ok) {                                                               \
+      TMA_CHECK(!relu || mp, "bn_backward: relu vec path needs the mask");     \
       at::Tensor psdy = at::empty({(int64_t)vg.row_blocks * C}, fopt);         \
       at::Tensor psdyx = at::empty({(int64_t)vg.row_blocks * C}, fopt);        \
       auto sargs = [&](auto kern) {                                            \
         hipLaunchKernelGGL(kern, dim3(vg.grid), dim3(256), 0, stream, xp, gp,  \
-                           yp, mean.data_ptr<float>(),                         \
+                           mp, mean.data_ptr<float>(),                         \
                            invstd.data_ptr<float>(), psdy.data_ptr<float>(),   \
                            psdyx.data_ptr<float>(), M, vg.nC8,                 \
                            vg.c8_per_block, vg.row_blocks);                    \
@@ -699,7 +713,7 @@ std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weig
                          sdy.data_ptr<float>(), sdyx.data_ptr<float>(), C);    \
       auto aargs = [&](auto kern) {                                            \
         hipLaunchKernelGGL(kern, dim3(vg.grid), dim3(256), 0, stream, xp, gp,  \
-                           yp, dxp, drp, mean.data_ptr<float>(),               \
+                           mp, dxp, drp, mean.data_ptr<float>(),               \
                            invstd.data_ptr<float>(), weight.data_ptr<float>(), \
                            sdy.data_ptr<float>(), sdyx.data_ptr<float>(),      \
                            invM, M, vg.nC8, vg.c8_per_block, vg.row_blocks);   \
@@ -709,6 +723,7 @@ std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weig
       else if (need_dres) aargs(bn_bwd_apply_vec<T, VEC, false, true>);        \
       else aargs(bn_bwd_apply_vec<T, VEC, false, false>);                      \
     } else {                                                                   \
+      TMA_CHECK(!relu || yp, "bn_backward: relu scalar path needs yout");      \
       sdy.zero_();                                                             \
       sdyx.zero_();                                                            \
       int c_blocks = (C + 63) / 64;                                            \
@@ -755,7 +770,7 @@ void register_batchnorm(pybind11::module_& m) {
   m.def("bn_backward", &bn_backward, py::arg("x"), py::arg("dy"),
         py::arg("weight"), py::arg("mean"), py::arg("invstd"),
         py::arg("yout") = py::none(), py::arg("relu") = false,
-        py::arg("need_dres") = false);
+        py::arg("need_dres") = false, py::arg("mask") = py::none());
 }
 
 }  // namespace tma

@@ -27,12 +27,14 @@ class _BnFunction(torch.autograd.Function):
         x = x.contiguous(memory_format=torch.channels_last)
         if res is not None:
             res = res.contiguous(memory_format=torch.channels_last)
-        y, mean, invstd = _C.bn_forward(x, weight, bias, running_mean,
-                                        running_var, training, momentum, eps,
-                                        relu, res)
-        # y is only needed for backward when the relu mask must be recovered.
+        y, mean, invstd, mask = _C.bn_forward(x, weight, bias, running_mean,
+                                              running_var, training, momentum,
+                                              eps, relu, res)
+        # Backward needs the relu mask: the 1-bit mask tensor on the vector
+        # path (1/16th of y), y itself on the scalar fallback / eval path.
+        need_y = relu and (mask.numel() == 0 or not training)
         ctx.save_for_backward(x, weight, mean, invstd,
-                              y if relu else torch.empty(0))
+                              y if need_y else torch.empty(0), mask)
         ctx.training = training
         ctx.relu = relu
         ctx.has_res = res is not None
@@ -40,7 +42,7 @@ class _BnFunction(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, weight, mean, invstd, y = ctx.saved_tensors
+        x, weight, mean, invstd, y, mask = ctx.saved_tensors
         dy = dy.contiguous(memory_format=torch.channels_last)
         if not ctx.training:
             # eval: stats are constants — no batch-statistic terms.
@@ -57,7 +59,9 @@ class _BnFunction(torch.autograd.Function):
             return (dx.contiguous(memory_format=torch.channels_last), dres,
                     dweight, dbias, None, None, None, None, None, None)
         out = _C.bn_backward(x, dy, weight, mean, invstd,
-                             y if ctx.relu else None, ctx.relu, ctx.has_res)
+                             y if (ctx.relu and y.numel()) else None,
+                             ctx.relu, ctx.has_res,
+                             mask if mask.numel() else None)
         dres = out[3] if ctx.has_res else None
         return (out[0], dres, out[1], out[2], None, None, None, None, None,
                 None)
