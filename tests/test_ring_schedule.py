@@ -128,3 +128,73 @@ def test_selector_mirrors_dispatch():
     # non-pow2 world: never RHD
     s = select_allreduce(torch.zeros(100), 3)
     assert s.algorithm == "ring"
+
+
+@pytest.mark.parametrize("world", [2, 3, 4, 8])
+def test_multiring_reduce_composition(world):
+    """The multi-ring GPU reduce (rccl_engine.hip): stripes split over the
+    rings; per ring, the RS schedule leaves position-of-rank-g owning
+    segment g of the stripe, which is then sent to root. Simulate the
+    stripe split + RS + gather and check the root reassembles the exact
+    elementwise sum while non-root inputs are preserved (scratch copy)."""
+    rings = ring_permutations(world)
+    nrings = len(rings)
+    count = 1000
+    # per-ring stripes: engine's split_aligned(count, nrings, align=4 fp32)
+    stripes = []
+    o = 0
+    per = (count + nrings - 1) // nrings
+    per = (per + 3) // 4 * 4
+    for r in range(nrings):
+        c = min(per, count - o) if o < count else 0
+        stripes.append((o, c))
+        o += c
+    assert sum(c for _, c in stripes) == count
+
+    # input: rank r's vector = [base + r for base in range(count)]
+    ranks = [[i + (r + 1) for i in range(count)] for r in range(world)]
+    expect = [i * world + world * (world + 1) // 2 for i in range(count)]
+    root = 1 % world
+    result = list(ranks[root])  # root reduces in place
+
+    for ring_idx, (off, cnt) in enumerate(stripes):
+        if cnt == 0:
+            continue
+        perm = rings[ring_idx]
+        # scratch copies for non-root (input preservation); root in place
+        work = [list(rk[off:off + cnt]) for rk in ranks]
+        # segment split within the stripe (split_segments semantics)
+        base, rem = divmod(cnt, world)
+        seg_off, seg_cnt = [], []
+        o2 = 0
+        for g in range(world):
+            c2 = base + (1 if g < rem else 0)
+            seg_off.append(o2)
+            seg_cnt.append(c2)
+            o2 += c2
+        # simulate ring RS on this stripe
+        rs = _C.ring_schedule(world, perm, 0)
+        for step in rs:
+            incoming = {}
+            for p in range(world):
+                send_seg, _ = step[p]
+                succ = perm[(p + 1) % world]
+                incoming[succ] = (send_seg,
+                                  work[perm[p]][seg_off[send_seg]:
+                                                seg_off[send_seg] +
+                                                seg_cnt[send_seg]])
+            for g, (seg, vals) in incoming.items():
+                for k, v in enumerate(vals):
+                    work[g][seg_off[seg] + k] += v
+        # gather: rank g sends its fully reduced segment g to root
+        for g in range(world):
+            if seg_cnt[g] == 0:
+                continue
+            src = work[g][seg_off[g]:seg_off[g] + seg_cnt[g]]
+            result[off + seg_off[g]:off + seg_off[g] + seg_cnt[g]] = src
+
+    assert result == expect, (world, result[:8], expect[:8])
+    # non-root inputs untouched (they reduced in scratch copies)
+    for r in range(world):
+        if r != root:
+            assert ranks[r] == [i + (r + 1) for i in range(count)]
