@@ -246,7 +246,19 @@ def main():
     synchronize_parameters(model)
     sync = GradientSynchronizer(model, bucket_mb=args.bucket_mb, average=True,
                                 comm_dtype=torch.bfloat16 if use_gpu else None)
-    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    # Optimizer elementwise work: ~214 small launches/step with the default
+    # per-param path (r3 trace). foreach batches them into multi-tensor ops;
+    # fused goes further (single kernel) where supported.
+    sgd_mode = os.environ.get("TORCHMPI_SGD", "foreach")
+    kw = {}
+    if sgd_mode == "fused":
+        kw["fused"] = True
+    elif sgd_mode == "foreach":
+        kw["foreach"] = True
+    try:
+        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9, **kw)
+    except (TypeError, RuntimeError):
+        opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
 
     wd = _Watchdog(float(os.environ.get("TORCHMPI_BENCH_STAGE_TIMEOUT", "600")))
     wd.arm("engine init + allreduce verify (first 8-rank contact)")
