@@ -249,7 +249,7 @@ def main():
     # Optimizer elementwise work: ~214 small launches/step with the default
     # per-param path (r3 trace). foreach batches them into multi-tensor ops;
     # fused goes further (single kernel) where supported.
-    sgd_mode = os.environ.get("TORCHMPI_SGD", "foreach")
+    sgd_mode = os.environ.get("TORCHMPI_SGD", "fused")
     kw = {}
     if sgd_mode == "fused":
         kw["fused"] = True

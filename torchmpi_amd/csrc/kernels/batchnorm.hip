@@ -18,6 +18,7 @@
 // Stats and affine params are fp32; x/y/dy may be bf16 or f32.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_cooperative_groups.h>
 
 #include <ATen/ATen.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
@@ -514,6 +515,345 @@ __global__ __launch_bounds__(256) void bn_bwd_apply_scalar(
   }
 }
 
+// ---- cooperative single-kernel BN (training fwd / bwd) --------------------
+// One hipLaunchCooperativeKernel replaces the stats -> finalize -> apply
+// 3-kernel chain: grid.sync() separates the phases, and the finalize runs
+// with the FULL grid's parallelism (the dedicated finalize kernel had only
+// C blocks and cost ~2.4 ms/step in launch/ramp — profiles/
+// resnet50_r2_fusion.md). Two-level finalize: slices of row_blocks reduce
+// into ps2/pq2 [nsl x C], then one thread per channel folds the slices.
+
+namespace cg = cooperative_groups;
+
+template <typename T, int VEC, bool RELU, bool RES>
+__global__ __launch_bounds__(256) void bn_fwd_coop_kernel(
+    const T* __restrict__ x, const T* __restrict__ res, T* __restrict__ y,
+    uint8_t* __restrict__ mask, float* __restrict__ psum,
+    float* __restrict__ psumsq, float* __restrict__ ps2,
+    float* __restrict__ pq2, float* __restrict__ mean,
+    float* __restrict__ invstd, float* __restrict__ running_mean,
+    float* __restrict__ running_var, const float* __restrict__ weight,
+    const float* __restrict__ bias, float momentum, float eps, size_t M,
+    int nC8, int c8_per_block, int row_blocks, int nsl) {
+  cg::grid_group grid = cg::this_grid();
+  const int c_blocks = nC8 / c8_per_block;
+  const int cb = blockIdx.x % c_blocks;
+  const int rb = blockIdx.x / c_blocks;
+  const int rows_per_iter = 256 / c8_per_block;
+  const int c8l = (int)threadIdx.x % c8_per_block;
+  const int rsub = (int)threadIdx.x / c8_per_block;
+  const int c8 = cb * c8_per_block + c8l;
+  const int C = nC8 * VEC;
+
+  // ---- phase 1: per-(row-block, channel) partial sums
+  {
+    float acc[VEC], accsq[VEC];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) acc[e] = accsq[e] = 0.f;
+    const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+    const T* base = x + (size_t)c8 * VEC;
+    for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+      T v[VEC];
+      *reinterpret_cast<uint4*>(v) =
+          *reinterpret_cast<const uint4*>(base + r * (size_t)C);
+#pragma unroll
+      for (int e = 0; e < VEC; ++e) {
+        float f = ldf(v[e]);
+        acc[e] += f;
+        accsq[e] += f * f;
+      }
+    }
+    __shared__ float ls[256 * VEC];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = acc[e];
+    __syncthreads();
+    if (rsub == 0) {
+      for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+        for (int e = 0; e < VEC; ++e)
+          acc[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = accsq[e];
+    __syncthreads();
+    if (rsub == 0) {
+      for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+        for (int e = 0; e < VEC; ++e)
+          accsq[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+      float* ps = psum + (size_t)rb * C + c8 * VEC;
+      float* pq = psumsq + (size_t)rb * C + c8 * VEC;
+#pragma unroll
+      for (int e = 0; e < VEC; e += 4) {
+        *reinterpret_cast<float4*>(ps + e) =
+            make_float4(acc[e], acc[e + 1], acc[e + 2], acc[e + 3]);
+        *reinterpret_cast<float4*>(pq + e) =
+            make_float4(accsq[e], accsq[e + 1], accsq[e + 2], accsq[e + 3]);
+      }
+    }
+  }
+  grid.sync();
+
+  // ---- phase 2a: slice-reduce partials (full grid parallelism)
+  {
+    const size_t T_ = (size_t)gridDim.x * 256;
+    const size_t gtid = (size_t)blockIdx.x * 256 + threadIdx.x;
+    const int span = (row_blocks + nsl - 1) / nsl;
+    for (size_t w = gtid; w < (size_t)C * nsl; w += T_) {
+      int c = (int)(w % C);
+      int s = (int)(w / C);
+      int r0 = s * span, r1 = min(r0 + span, row_blocks);
+      float a = 0.f, b = 0.f;
+      for (int r = r0; r < r1; ++r) {
+        a += psum[(size_t)r * C + c];
+        b += psumsq[(size_t)r * C + c];
+      }
+      ps2[(size_t)s * C + c] = a;
+      pq2[(size_t)s * C + c] = b;
+    }
+  }
+  grid.sync();
+
+  // ---- phase 2b: per-channel mean/invstd (+ running stats)
+  {
+    const size_t T_ = (size_t)gridDim.x * 256;
+    const size_t gtid = (size_t)blockIdx.x * 256 + threadIdx.x;
+    const float Mf = (float)M;
+    for (size_t c = gtid; c < (size_t)C; c += T_) {
+      float s = 0.f, ss = 0.f;
+      for (int k = 0; k < nsl; ++k) {
+        s += ps2[(size_t)k * C + c];
+        ss += pq2[(size_t)k * C + c];
+      }
+      float m = s / Mf;
+      float var = fmaxf(ss / Mf - m * m, 0.f);
+      mean[c] = m;
+      invstd[c] = rsqrtf(var + eps);
+      if (running_mean) {
+        running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * m;
+        float unbiased = Mf > 1.f ? var * Mf / (Mf - 1.f) : var;
+        running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+      }
+    }
+  }
+  grid.sync();
+
+  // ---- phase 3: apply (y = [relu](sc*x + sh [+ res]); mask bits)
+  {
+    float sc[VEC], sh[VEC];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      int c = c8 * VEC + e;
+      sc[e] = weight[c] * invstd[c];
+      sh[e] = bias[c] - mean[c] * sc[e];
+    }
+    const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+    const T* xb = x + (size_t)c8 * VEC;
+    const T* rb2 = RES ? res + (size_t)c8 * VEC : nullptr;
+    T* yb = y + (size_t)c8 * VEC;
+    for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+      T v[VEC], o[VEC], rv[VEC];
+      *reinterpret_cast<uint4*>(v) =
+          *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
+      if (RES)
+        *reinterpret_cast<uint4*>(rv) =
+            *reinterpret_cast<const uint4*>(rb2 + r * (size_t)C);
+      unsigned mb = 0;
+#pragma unroll
+      for (int e = 0; e < VEC; ++e) {
+        float f = ldf(v[e]) * sc[e] + sh[e];
+        if (RES) f += ldf(rv[e]);
+        if (RELU) {
+          mb |= (unsigned)(f > 0.f) << e;
+          f = fmaxf(f, 0.f);
+        }
+        o[e] = stf<T>(f);
+      }
+      *reinterpret_cast<uint4*>(yb + r * (size_t)C) = *reinterpret_cast<uint4*>(o);
+      if (RELU) mask[r * (size_t)nC8 + c8] = (uint8_t)mb;
+    }
+  }
+}
+
+template <typename T, int VEC, bool RELU, bool RES>
+__global__ __launch_bounds__(256) void bn_bwd_coop_kernel(
+    const T* __restrict__ x, const T* __restrict__ dy,
+    const uint8_t* __restrict__ mask, T* __restrict__ dx,
+    T* __restrict__ dres, float* __restrict__ psdy, float* __restrict__ psdyx,
+    float* __restrict__ ps2, float* __restrict__ pq2,
+    float* __restrict__ sdy, float* __restrict__ sdyx,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    const float* __restrict__ weight, float invM, size_t M, int nC8,
+    int c8_per_block, int row_blocks, int nsl) {
+  cg::grid_group grid = cg::this_grid();
+  const int c_blocks = nC8 / c8_per_block;
+  const int cb = blockIdx.x % c_blocks;
+  const int rb = blockIdx.x / c_blocks;
+  const int rows_per_iter = 256 / c8_per_block;
+  const int c8l = (int)threadIdx.x % c8_per_block;
+  const int rsub = (int)threadIdx.x / c8_per_block;
+  const int c8 = cb * c8_per_block + c8l;
+  const int C = nC8 * VEC;
+
+  // ---- phase 1: partial sdy / sdyx (relu-masked)
+  {
+    float mu[VEC], is[VEC], acc[VEC], accx[VEC];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      mu[e] = mean[c8 * VEC + e];
+      is[e] = invstd[c8 * VEC + e];
+      acc[e] = accx[e] = 0.f;
+    }
+    const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+    const T* xb = x + (size_t)c8 * VEC;
+    const T* gb = dy + (size_t)c8 * VEC;
+    for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+      T xv[VEC], gv[VEC];
+      *reinterpret_cast<uint4*>(xv) =
+          *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
+      *reinterpret_cast<uint4*>(gv) =
+          *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
+      unsigned mb = RELU ? mask[r * (size_t)nC8 + c8] : 0u;
+#pragma unroll
+      for (int e = 0; e < VEC; ++e) {
+        float g = ldf(gv[e]);
+        if (RELU && !((mb >> e) & 1u)) g = 0.f;
+        float xh = (ldf(xv[e]) - mu[e]) * is[e];
+        acc[e] += g;
+        accx[e] += g * xh;
+      }
+    }
+    __shared__ float ls[256 * VEC];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = acc[e];
+    __syncthreads();
+    if (rsub == 0) {
+      for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+        for (int e = 0; e < VEC; ++e)
+          acc[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) ls[threadIdx.x * VEC + e] = accx[e];
+    __syncthreads();
+    if (rsub == 0) {
+      for (int w = 1; w < rows_per_iter; ++w)
+#pragma unroll
+        for (int e = 0; e < VEC; ++e)
+          accx[e] += ls[(w * c8_per_block + c8l) * VEC + e];
+      float* pa = psdy + (size_t)rb * C + c8 * VEC;
+      float* pb = psdyx + (size_t)rb * C + c8 * VEC;
+#pragma unroll
+      for (int e = 0; e < VEC; e += 4) {
+        *reinterpret_cast<float4*>(pa + e) =
+            make_float4(acc[e], acc[e + 1], acc[e + 2], acc[e + 3]);
+        *reinterpret_cast<float4*>(pb + e) =
+            make_float4(accx[e], accx[e + 1], accx[e + 2], accx[e + 3]);
+      }
+    }
+  }
+  grid.sync();
+
+  // ---- phase 2a: slice reduce
+  {
+    const size_t T_ = (size_t)gridDim.x * 256;
+    const size_t gtid = (size_t)blockIdx.x * 256 + threadIdx.x;
+    const int span = (row_blocks + nsl - 1) / nsl;
+    for (size_t w = gtid; w < (size_t)C * nsl; w += T_) {
+      int c = (int)(w % C);
+      int s = (int)(w / C);
+      int r0 = s * span, r1 = min(r0 + span, row_blocks);
+      float a = 0.f, b = 0.f;
+      for (int r = r0; r < r1; ++r) {
+        a += psdy[(size_t)r * C + c];
+        b += psdyx[(size_t)r * C + c];
+      }
+      ps2[(size_t)s * C + c] = a;
+      pq2[(size_t)s * C + c] = b;
+    }
+  }
+  grid.sync();
+
+  // ---- phase 2b: final per-channel sums
+  {
+    const size_t T_ = (size_t)gridDim.x * 256;
+    const size_t gtid = (size_t)blockIdx.x * 256 + threadIdx.x;
+    for (size_t c = gtid; c < (size_t)C; c += T_) {
+      float a = 0.f, b = 0.f;
+      for (int k = 0; k < nsl; ++k) {
+        a += ps2[(size_t)k * C + c];
+        b += pq2[(size_t)k * C + c];
+      }
+      sdy[c] = a;
+      sdyx[c] = b;
+    }
+  }
+  grid.sync();
+
+  // ---- phase 3: apply (dx = A*g + B*x + D; dres = g)
+  {
+    float A[VEC], B[VEC], D[VEC];
+#pragma unroll
+    for (int e = 0; e < VEC; ++e) {
+      int c = c8 * VEC + e;
+      float a = weight[c] * invstd[c];
+      float b = -a * invstd[c] * sdyx[c] * invM;
+      A[e] = a;
+      B[e] = b;
+      D[e] = -a * sdy[c] * invM - b * mean[c];
+    }
+    const size_t row_stride = (size_t)row_blocks * rows_per_iter;
+    const T* xb = x + (size_t)c8 * VEC;
+    const T* gb = dy + (size_t)c8 * VEC;
+    T* ob = dx + (size_t)c8 * VEC;
+    T* eb = RES ? dres + (size_t)c8 * VEC : nullptr;
+    for (size_t r = (size_t)rb * rows_per_iter + rsub; r < M; r += row_stride) {
+      T xv[VEC], gv[VEC], o[VEC], ev[VEC];
+      *reinterpret_cast<uint4*>(xv) =
+          *reinterpret_cast<const uint4*>(xb + r * (size_t)C);
+      *reinterpret_cast<uint4*>(gv) =
+          *reinterpret_cast<const uint4*>(gb + r * (size_t)C);
+      unsigned mb = RELU ? mask[r * (size_t)nC8 + c8] : 0u;
+#pragma unroll
+      for (int e = 0; e < VEC; ++e) {
+        float g = ldf(gv[e]);
+        if (RELU && !((mb >> e) & 1u)) g = 0.f;
+        o[e] = stf<T>(A[e] * g + B[e] * ldf(xv[e]) + D[e]);
+        if (RES) ev[e] = stf<T>(g);
+      }
+      *reinterpret_cast<uint4*>(ob + r * (size_t)C) = *reinterpret_cast<uint4*>(o);
+      if (RES)
+        *reinterpret_cast<uint4*>(eb + r * (size_t)C) =
+            *reinterpret_cast<uint4*>(ev);
+    }
+  }
+}
+
+// Max co-resident blocks for a cooperative kernel (cached per instantiation);
+// -1 = cooperative launch unavailable.
+template <typename KernT>
+int coop_max_blocks(KernT kern) {
+  static int cached = -2;
+  if (cached == -2) {
+    const char* off = std::getenv("TORCHMPI_BN_COOP");
+    if (off && std::string(off) == "0") {
+      cached = -1;
+      return cached;
+    }
+    int dev = 0;
+    if (hipGetDevice(&dev) != hipSuccess) return (cached = -1);
+    hipDeviceProp_t prop;
+    if (hipGetDeviceProperties(&prop, dev) != hipSuccess) return (cached = -1);
+    if (!prop.cooperativeLaunch) return (cached = -1);
+    int nb = 0;
+    hipError_t e = hipOccupancyMaxActiveBlocksPerMultiprocessor(&nb, kern, 256, 0);
+    cached = (e == hipSuccess && nb > 0) ? nb * prop.multiProcessorCount : -1;
+  }
+  return cached;
+}
+
 void check_bn_input(const at::Tensor& x) {
   TMA_CHECK(x.is_cuda() && x.dim() == 4, "bn: 4D CUDA tensor required");
   TMA_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
@@ -552,6 +892,79 @@ std::vector<at::Tensor> bn_forward(at::Tensor x, at::Tensor weight, at::Tensor b
   VecGeom vg = vec_geom(C, vecw, M);
   auto fopt = at::TensorOptions().dtype(at::kFloat).device(x.device());
   at::Tensor mean, invstd;
+
+  // Cooperative single-kernel fast path: stats -> finalize -> apply in one
+  // launch with grid.sync (falls back to the 3-kernel chain whenever
+  // cooperative launch is unavailable or the grid exceeds residency).
+  if (training && vg.ok) {
+    mean = at::empty({C}, fopt);
+    invstd = at::empty({C}, fopt);
+    at::Tensor y = at::empty_like(x);
+    at::Tensor mask;
+    if (relu)
+      mask = at::empty({(int64_t)(M * (size_t)vg.nC8)},
+                       at::TensorOptions().dtype(at::kByte).device(x.device()));
+    bool done = false;
+#define TMA_BN_FWD_COOP(T, VEC)                                                \
+  do {                                                                         \
+    auto try_coop = [&](auto kern) -> bool {                                   \
+      int maxb = coop_max_blocks(kern);                                        \
+      if (maxb <= 0 || vg.c_blocks > maxb) return false;                       \
+      int rbk = std::min(vg.row_blocks, std::max(1, maxb / vg.c_blocks));      \
+      int grid = vg.c_blocks * rbk;                                            \
+      int nsl = std::min(64, rbk);                                             \
+      at::Tensor psum = at::empty({(int64_t)rbk * C}, fopt);                   \
+      at::Tensor psumsq = at::empty({(int64_t)rbk * C}, fopt);                 \
+      at::Tensor ps2 = at::empty({(int64_t)nsl * C}, fopt);                    \
+      at::Tensor pq2 = at::empty({(int64_t)nsl * C}, fopt);                    \
+      const T* xp = (const T*)x.data_ptr();                                    \
+      const T* rp = has_res ? (const T*)res.data_ptr() : nullptr;              \
+      T* yp = (T*)y.data_ptr();                                                \
+      uint8_t* mkp = mask.defined() ? mask.data_ptr<uint8_t>() : nullptr;      \
+      float* psp = psum.data_ptr<float>();                                     \
+      float* pqp = psumsq.data_ptr<float>();                                   \
+      float* p2 = ps2.data_ptr<float>();                                       \
+      float* q2 = pq2.data_ptr<float>();                                       \
+      float* mnp = mean.data_ptr<float>();                                     \
+      float* ivp = invstd.data_ptr<float>();                                   \
+      float* rmp = running_mean.defined() ? running_mean.data_ptr<float>()     \
+                                          : nullptr;                          \
+      float* rvp = running_var.defined() ? running_var.data_ptr<float>()       \
+                                         : nullptr;                          \
+      const float* wp = weight.data_ptr<float>();                             \
+      const float* bp = bias.data_ptr<float>();                               \
+      float mom = (float)momentum, ep = (float)eps;                           \
+      size_t Mv = M;                                                          \
+      int nc8 = vg.nC8, c8pb = vg.c8_per_block;                               \
+      void* kargs[] = {&xp, &rp, &yp, &mkp, &psp, &pqp, &p2,  &q2,            \
+                       &mnp, &ivp, &rmp, &rvp, &wp, &bp, &mom, &ep,           \
+                       &Mv, &nc8, &c8pb, &rbk, &nsl};                         \
+      hipError_t e = hipLaunchCooperativeKernel(                              \
+          reinterpret_cast<const void*>(kern), dim3(grid), dim3(256), kargs,  \
+          0, stream);                                                         \
+      if (e != hipSuccess) {                                                  \
+        (void)hipGetLastError();                                              \
+        return false;                                                         \
+      }                                                                       \
+      return true;                                                            \
+    };                                                                        \
+    if (relu && has_res) done = try_coop(bn_fwd_coop_kernel<T, VEC, true, true>);   \
+    else if (relu) done = try_coop(bn_fwd_coop_kernel<T, VEC, true, false>);  \
+    else if (has_res) done = try_coop(bn_fwd_coop_kernel<T, VEC, false, true>);     \
+    else done = try_coop(bn_fwd_coop_kernel<T, VEC, false, false>);           \
+  } while (0)
+    if (bf16)
+      TMA_BN_FWD_COOP(__hip_bfloat16, 8);
+    else
+      TMA_BN_FWD_COOP(float, 4);
+#undef TMA_BN_FWD_COOP
+    if (done) {
+      if (!mask.defined())
+        mask = at::empty({0},
+                         at::TensorOptions().dtype(at::kByte).device(x.device()));
+      return {y, mean, invstd, mask};
+    }
+  }
 
   if (training) {
     mean = at::empty({C}, fopt);
@@ -683,6 +1096,69 @@ std::vector<at::Tensor> bn_backward(at::Tensor x, at::Tensor dy, at::Tensor weig
   at::Tensor dres;
   if (need_dres) dres = at::empty_like(x);
   float invM = 1.f / (float)M;
+
+  // Cooperative single-kernel fast path (mirrors forward).
+  if (vg.ok && (!relu || mask.defined())) {
+    bool done = false;
+#define TMA_BN_BWD_COOP(T, VEC)                                                \
+  do {                                                                         \
+    auto try_coop = [&](auto kern) -> bool {                                   \
+      int maxb = coop_max_blocks(kern);                                        \
+      if (maxb <= 0 || vg.c_blocks > maxb) return false;                       \
+      int rbk = std::min(vg.row_blocks, std::max(1, maxb / vg.c_blocks));      \
+      int grid = vg.c_blocks * rbk;                                            \
+      int nsl = std::min(64, rbk);                                             \
+      at::Tensor psdy = at::empty({(int64_t)rbk * C}, fopt);                   \
+      at::Tensor psdyx = at::empty({(int64_t)rbk * C}, fopt);                  \
+      at::Tensor ps2 = at::empty({(int64_t)nsl * C}, fopt);                    \
+      at::Tensor pq2 = at::empty({(int64_t)nsl * C}, fopt);                    \
+      const T* xp = (const T*)x.data_ptr();                                    \
+      const T* gp = (const T*)dy.data_ptr();                                   \
+      const uint8_t* mkp =                                                     \
+          mask.defined() ? mask.data_ptr<uint8_t>() : nullptr;                 \
+      T* dxp = (T*)dx.data_ptr();                                              \
+      T* drp = need_dres ? (T*)dres.data_ptr() : nullptr;                      \
+      float* pa = psdy.data_ptr<float>();                                      \
+      float* pb = psdyx.data_ptr<float>();                                     \
+      float* p2 = ps2.data_ptr<float>();                                       \
+      float* q2 = pq2.data_ptr<float>();                                       \
+      float* sa = sdy.data_ptr<float>();                                       \
+      float* sb = sdyx.data_ptr<float>();                                      \
+      const float* mnp = mean.data_ptr<float>();                               \
+      const float* ivp = invstd.data_ptr<float>();                             \
+      const float* wp = weight.data_ptr<float>();                              \
+      float im = invM;                                                         \
+      size_t Mv = M;                                                           \
+      int nc8 = vg.nC8, c8pb = vg.c8_per_block;                                \
+      void* kargs[] = {&xp, &gp, &mkp, &dxp, &drp, &pa, &pb, &p2, &q2,         \
+                       &sa, &sb, &mnp, &ivp, &wp, &im, &Mv, &nc8, &c8pb,       \
+                       &rbk, &nsl};                                            \
+      hipError_t e = hipLaunchCooperativeKernel(                               \
+          reinterpret_cast<const void*>(kern), dim3(grid), dim3(256), kargs,   \
+          0, stream);                                                          \
+      if (e != hipSuccess) {                                                   \
+        (void)hipGetLastError();                                               \
+        return false;                                                          \
+      }                                                                        \
+      return true;                                                             \
+    };                                                                         \
+    if (relu && need_dres)                                                     \
+      done = try_coop(bn_bwd_coop_kernel<T, VEC, true, true>);                 \
+    else if (relu) done = try_coop(bn_bwd_coop_kernel<T, VEC, true, false>);   \
+    else if (need_dres)                                                        \
+      done = try_coop(bn_bwd_coop_kernel<T, VEC, false, true>);                \
+    else done = try_coop(bn_bwd_coop_kernel<T, VEC, false, false>);            \
+  } while (0)
+    if (bf16)
+      TMA_BN_BWD_COOP(__hip_bfloat16, 8);
+    else
+      TMA_BN_BWD_COOP(float, 4);
+#undef TMA_BN_BWD_COOP
+    if (done) {
+      if (need_dres) return {dx, sdyx, sdy, dres};
+      return {dx, sdyx, sdy};
+    }
+  }
 
 #define TMA_BN_BWD(T, VEC)                                                     \
   do {                                                                         \
