@@ -837,8 +837,12 @@ template <typename KernT>
 int coop_max_blocks(KernT kern) {
   static int cached = -2;
   if (cached == -2) {
-    const char* off = std::getenv("TORCHMPI_BN_COOP");
-    if (off && std::string(off) == "0") {
+    // MEASURED OFF by default: on ROCm 7.2 the cooperative launch path cost
+    // far more than the finalize it fuses away (bn_bench 0.2-0.4x, end-to-end
+    // 8099 -> 3121 samples/s; gpurun_out/r7_*). Numerics verified; opt in
+    // with TORCHMPI_BN_COOP=1 to re-evaluate on newer stacks.
+    const char* on = std::getenv("TORCHMPI_BN_COOP");
+    if (!(on && std::string(on) == "1")) {
       cached = -1;
       return cached;
     }
