@@ -17,6 +17,7 @@
 //     row_blocks (=2048) and cost a flat ~420 us regardless of size.
 // Stats and affine params are fp32; x/y/dy may be bf16 or f32.
 #include <hip/hip_runtime.h>
+#include <cstdlib>
 #include <hip/hip_bf16.h>
 #include <hip/hip_cooperative_groups.h>
 
@@ -63,6 +64,15 @@ struct VecGeom {
   int nC8 = 0, c8_per_block = 0, c_blocks = 0, row_blocks = 0, grid = 0;
 };
 
+inline int bn_rb_cap() {
+  static int cap = -1;
+  if (cap < 0) {
+    const char* v = std::getenv("TORCHMPI_BN_RB");
+    cap = v ? std::max(64, atoi(v)) : 4096;  // measured default, see profiles/
+  }
+  return cap;
+}
+
 VecGeom vec_geom(int C, int VEC, size_t M) {
   VecGeom g;
   if (C % VEC != 0) return g;
@@ -76,7 +86,7 @@ VecGeom vec_geom(int C, int VEC, size_t M) {
   g.c_blocks = g.nC8 / g.c8_per_block;
   int rows_per_iter = 256 / g.c8_per_block;
   int max_rb = (int)((M + rows_per_iter - 1) / rows_per_iter);
-  g.row_blocks = std::min(std::max(4096 / g.c_blocks, 1), std::max(max_rb, 1));
+  g.row_blocks = std::min(std::max(bn_rb_cap() / g.c_blocks, 1), std::max(max_rb, 1));
   // Small-M/large-C shapes: keep the partial buffers (row_blocks * C * 8 B)
   // under half the data read, or the reduction traffic dominates.
   int traffic_cap = std::max(1, (int)(M / VEC));
