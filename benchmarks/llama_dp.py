@@ -14,20 +14,6 @@ import os
 import sys
 import time
 
-# hipBLASLt/rocBLAS GEMM autotuning: preload the committed gfx950 table
-# (read-only) unless the caller configured TunableOp explicitly. Tuned on
-# the llama3_8b mb4/seq4096 shapes; validator mismatch => safely ignored.
-# TunableOp resolves "<name>.csv" to "<name><device>.csv" on read/write,
-# so the committed table carries the trailing 0 and FILENAME omits it.
-_TOP_TABLE = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                          "tunableop_gfx950_llama.csv")
-if (os.environ.get("TORCHMPI_TUNABLEOP", "1") != "0"
-        and "PYTORCH_TUNABLEOP_ENABLED" not in os.environ
-        and os.path.exists(_TOP_TABLE.replace(".csv", "0.csv"))):
-    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
-    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
-    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = _TOP_TABLE
-
 import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
