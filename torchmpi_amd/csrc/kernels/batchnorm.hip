@@ -68,7 +68,7 @@ inline int bn_rb_cap() {
   static int cap = -1;
   if (cap < 0) {
     const char* v = std::getenv("TORCHMPI_BN_RB");
-    cap = v ? std::max(64, atoi(v)) : 4096;  // measured default, see profiles/
+    cap = v ? std::max(64, atoi(v)) : 512;  // measured sweep, see profiles/
   }
   return cap;
 }
