@@ -496,20 +496,25 @@ class RcclEngine {
     // root reduces in place: its tensor becomes the result anyway.
     char* tbase = static_cast<char*>(t.data_ptr());
     char* base = tbase;
+    const size_t align = 16 / es;
+    int nrings = count * es >= multiring_min_bytes_ ? used_rings() : 1;
     at::Tensor scratch;
     if (rank_ != root) {
       scratch = scratch_like(count * es, s0);
+      // The scratch copy is READ by every ring stream below — record each
+      // so the caching allocator cannot reuse it early.
+      for (int r = 1; r < nrings; ++r)
+        record_stream_uses(scratch, streams_[r]);
       work->keep.push_back(scratch);
       TMA_HIP_CHECK(hipMemcpyAsync(scratch.data_ptr(), tbase, count * es,
                                    hipMemcpyDeviceToDevice, s0.stream()));
       base = static_cast<char*>(scratch.data_ptr());
     }
+    record_stream_uses(t, s0);  // s0's memcpy/recv reads t
 
     // Multi-ring (same striping as allreduce): ring r reduce-scatters its
     // stripe then gathers the stripe's segments to root over comms_[r],
     // using all xGMI links instead of one.
-    const size_t align = 16 / es;
-    int nrings = count * es >= multiring_min_bytes_ ? used_rings() : 1;
     std::vector<size_t> soff, scnt;
     split_aligned(count, nrings, align, soff, scnt);
     hipEvent_t fork = nullptr;
