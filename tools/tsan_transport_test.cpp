@@ -113,6 +113,33 @@ int main() {
       printf("PS mismatch: %f vs %f\n", got[0], 24.f * W);
       failures++;
     }
+    // Phase 2b: mutual large fetch — every rank receives the full tensor at
+    // once while shards (4 MB each) far exceed the socket buffers. Drives
+    // the loop's non-blocking out-queue drain + incremental frame reads
+    // (the round-1 advisor deadlock, both directions) under TSAN.
+    const size_t BIGN = (size_t)4 << 20;  // 16 MB fp32
+    {
+      std::vector<float> init(BIGN, 3.0f);
+      for (int r = 0; r < W; ++r)
+        pss[r]->register_tensor(8, BIGN, DType::F32, init.data());
+      std::vector<std::thread> ts2;
+      std::vector<int> bad(W, 0);
+      for (int r = 0; r < W; ++r) {
+        ts2.emplace_back([&, r] {
+          std::vector<float> out(BIGN);
+          pss[r]->receive(8, out.data())->wait();
+          if (std::fabs(out[0] - 3.0f) > 1e-6 ||
+              std::fabs(out[BIGN - 1] - 3.0f) > 1e-6)
+            bad[r] = 1;
+        });
+      }
+      for (auto& t : ts2) t.join();
+      for (int r = 0; r < W; ++r)
+        if (bad[r]) {
+          printf("PS big fetch mismatch at rank %d\n", r);
+          failures++;
+        }
+    }
     for (int r = 0; r < W; ++r) pss[r]->stop();
   }
 
