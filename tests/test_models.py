@@ -57,3 +57,29 @@ def test_llama3_8b_param_count():
     total = (c.vocab_size * c.dim + c.n_layers * per_layer + c.dim
              + c.dim * c.vocab_size)
     assert abs(total - 8.03e9) < 0.01e9, total
+
+
+def test_llama_block_pair_composition_cpu():
+    """The (residual, pending-branch) Block pipeline must equal the textbook
+    x + attn(ln1(x)); x + mlp(ln2(x)) composition (fp32, CPU eager path)."""
+    import torch
+    from torchmpi_amd.models.llama import (Block, _add_norm, _rope_freqs,
+                                           llama_debug)
+    torch.manual_seed(5)
+    cfg = llama_debug()
+    blk1, blk2 = Block(cfg), Block(cfg)
+    freqs = _rope_freqs(cfg.dim // cfg.n_heads, 32, cfg.rope_theta,
+                        torch.device("cpu"))
+    x = torch.randn(2, 8, cfg.dim)
+    with torch.no_grad():
+        # pair pipeline through two blocks + final norm-style fold
+        res, branch = blk1(x, None, freqs)
+        res, branch = blk2(res, branch, freqs)
+        out_pair = res + branch
+        # textbook composition
+        ref = x
+        for blk in (blk1, blk2):
+            ref = ref + blk.attn(blk.ln1(ref), freqs)
+            ref = ref + blk.mlp(blk.ln2(ref))
+    assert torch.allclose(out_pair, ref, atol=1e-5, rtol=1e-5), \
+        (out_pair - ref).abs().max()
