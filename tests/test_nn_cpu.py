@@ -220,3 +220,46 @@ def _bf16_comm_dtype_body(tm):
 
 def test_bf16_comm_dtype():
     run_distributed(_bf16_comm_dtype_body, world=2)
+
+
+def _llama_dp_equivalence_body(tm):
+    """Flagship-transformer DP step equivalence: W-rank training with
+    averaged grads == single-process training on the concatenated batch
+    (reference stack 3.5, applied to the Llama model family)."""
+    from torchmpi_amd.models import LlamaModel, llama_debug
+    from torchmpi_amd.nn import GradientSynchronizer, synchronize_parameters
+    W, r = tm.size(), tm.rank()
+    cfg = llama_debug()
+    cfg.n_layers = 2
+    torch.manual_seed(11)
+    model = LlamaModel(cfg)
+    synchronize_parameters(model)
+    sync = GradientSynchronizer(model, average=True)
+    opt = torch.optim.SGD(model.parameters(), lr=0.05)
+
+    torch.manual_seed(11)
+    ref = LlamaModel(cfg)
+    ref_opt = torch.optim.SGD(ref.parameters(), lr=0.05)
+
+    for step in range(2):
+        torch.manual_seed(300 + step)
+        toks = torch.randint(0, cfg.vocab_size, (W, 2, 32))
+        tgts = torch.randint(0, cfg.vocab_size, (W, 2, 32))
+        opt.zero_grad(set_to_none=False)
+        loss = model(toks[r], tgts[r])
+        loss.backward()
+        sync.finish()
+        opt.step()
+
+        ref_opt.zero_grad()
+        ref_loss = sum(ref(toks[k], tgts[k]) for k in range(W)) / W
+        ref_loss.backward()
+        ref_opt.step()
+
+    for p, q in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.data, q.data, atol=1e-5, rtol=1e-4), \
+            (p.shape, (p.data - q.data).abs().max())
+
+
+def test_llama_dp_equivalence():
+    run_distributed(_llama_dp_equivalence_body, world=2)
