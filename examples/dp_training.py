@@ -92,6 +92,15 @@ def _spawn(rank, world, port, checkpoint=None, resume=False):
     main_worker(checkpoint, resume)
 
 
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
 if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--world", type=int, default=0,
@@ -101,7 +110,7 @@ if __name__ == "__main__":
     args = ap.parse_args()
     if args.world > 0:
         import torch.multiprocessing as mp
-        mp.spawn(_spawn, (args.world, 29610, args.checkpoint, args.resume),
+        mp.spawn(_spawn, (args.world, _free_port(), args.checkpoint, args.resume),
                  nprocs=args.world, join=True)
     else:
         main_worker(args.checkpoint, args.resume)

@@ -69,9 +69,18 @@ def _spawn(rank, world, port):
     main_worker()
 
 
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
 if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--world", type=int, default=2)
     args = ap.parse_args()
     import torch.multiprocessing as mp
-    mp.spawn(_spawn, (args.world, 29612), nprocs=args.world, join=True)
+    mp.spawn(_spawn, (args.world, _free_port()), nprocs=args.world, join=True)
