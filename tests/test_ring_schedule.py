@@ -198,3 +198,44 @@ def test_multiring_reduce_composition(world):
     for r in range(world):
         if r != root:
             assert ranks[r] == [i + (r + 1) for i in range(count)]
+
+
+# -- property-based invariants (hypothesis) ---------------------------------
+try:
+    from hypothesis import given, settings, strategies as st
+
+    @given(world=st.integers(2, 12), seed=st.integers(0, 10**6))
+    @settings(max_examples=60, deadline=None)
+    def test_ring_schedule_any_permutation(world, seed):
+        """The RS/AG schedule invariants hold for ARBITRARY ring
+        permutations, not just the engine's generated ones (the engine
+        claims permutation-independence in rccl_engine.hip)."""
+        import random
+        rng = random.Random(seed)
+        perm = list(range(world))
+        rng.shuffle(perm)
+        simulate_allreduce(world, perm)
+
+    @given(count=st.integers(0, 100000), parts=st.integers(1, 16),
+           align=st.sampled_from([1, 2, 4, 8]))
+    @settings(max_examples=100, deadline=None)
+    def test_segment_parts_any_shape(count, parts, align):
+        """split/segment tables: disjoint, complete, aligned starts."""
+        off, cnt = _C.segment_parts([0], [count], 0, 1, align)
+        assert off == [0] and cnt == [count]
+        # multi-part split over a single global segment
+        total = 0
+        prev_end = 0
+        for r in range(parts):
+            off_r, cnt_r = _C.segment_parts([0], [count], r, parts, align)
+            assert off_r[0] >= prev_end or cnt_r[0] == 0
+            if cnt_r[0] > 0:
+                assert off_r[0] == prev_end
+                if r < parts - 1 and off_r[0] + cnt_r[0] < count:
+                    assert (off_r[0] + cnt_r[0]) % align == 0 or \
+                        off_r[0] + cnt_r[0] == count
+                prev_end = off_r[0] + cnt_r[0]
+            total += cnt_r[0]
+        assert total == count
+except ImportError:  # pragma: no cover
+    pass
