@@ -243,7 +243,12 @@ def main():
             convert_batchnorm(model)  # our gfx950 NHWC BN kernels
         model = model.to(memory_format=torch.channels_last)
         torch.backends.cudnn.benchmark = True  # MIOpen algo search in warmup
+    # First RCCL contact is HERE (161 parameter broadcasts over the new
+    # topology) — the watchdog must already be armed.
+    wd = _Watchdog(float(os.environ.get("TORCHMPI_BENCH_STAGE_TIMEOUT", "600")))
+    wd.arm("engine init + parameter broadcast (first RCCL contact)")
     synchronize_parameters(model)
+    wd.disarm()
     sync = GradientSynchronizer(model, bucket_mb=args.bucket_mb, average=True,
                                 comm_dtype=torch.bfloat16 if use_gpu else None)
     # Optimizer elementwise work: ~214 small launches/step with the default
@@ -260,8 +265,7 @@ def main():
     except (TypeError, RuntimeError):
         opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
 
-    wd = _Watchdog(float(os.environ.get("TORCHMPI_BENCH_STAGE_TIMEOUT", "600")))
-    wd.arm("engine init + allreduce verify (first 8-rank contact)")
+    wd.arm("allreduce verify ladder")
     graph_status = verify_allreduce(tm, device)
     wd.disarm()
 
