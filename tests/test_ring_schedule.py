@@ -239,3 +239,40 @@ try:
         assert total == count
 except ImportError:  # pragma: no cover
     pass
+
+
+@pytest.mark.parametrize("world", [2, 3, 4, 8])
+@pytest.mark.parametrize("nchunks", [1, 2, 5])
+def test_chain_broadcast_schedule(world, nchunks):
+    """Mirror of RcclEngine::chain_broadcast's per-iteration schedule
+    (rccl_engine.hip): at iteration c, position p>0 receives chunk c
+    (c < nchunks) and position p<W-1 sends chunk c-1 (c > 0). Verifies
+    (a) per-(pair, direction) op sequences match in order (RCCL grouped
+    p2p matches by call order, not tags), and (b) every rank holds every
+    chunk with stream-order dependencies respected."""
+    for perm in ring_permutations(world):
+        for rootpos in range(world):
+            root = perm[rootpos]
+            # op sequences per directed pair, and per-rank chunk arrival order
+            sends = {}  # (src, dst) -> [chunk,...]
+            recvs = {}
+            for p in range(world):
+                mypos = (p - rootpos + world) % world
+                nxt = perm[(p + 1) % world]
+                prv = perm[(p - 1 + world) % world]
+                me = perm[p]
+                for c in range(nchunks + 1):
+                    if mypos > 0 and c < nchunks:
+                        recvs.setdefault((prv, me), []).append(c)
+                    if mypos < world - 1 and c > 0:
+                        sends.setdefault((me, nxt), []).append(c - 1)
+            assert sends.keys() == recvs.keys()
+            for pair in sends:
+                assert sends[pair] == recvs[pair], (pair, sends[pair],
+                                                    recvs[pair])
+            # dependency: a rank sends chunk j only at iteration j+1, after
+            # receiving it at iteration j (same stream => ordered); the root
+            # owns all chunks from the start. Therefore every non-root rank
+            # receives chunks 0..nchunks-1 exactly once => full delivery.
+            for (src, dst), chunks in recvs.items():
+                assert chunks == list(range(nchunks))
