@@ -128,3 +128,22 @@ def _hier_2x4_body(tm):
 
 def test_hierarchical_2x4():
     run_distributed(_hier_2x4_body, world=8)
+
+
+def _push_string_key_body(tm):
+    """Reference parity: TorchMPI pushed communicators by STRING key; all
+    ranks pushing the same string land in one subgroup."""
+    W, r = tm.size(), tm.rank()
+    sub = tm.push_communicator("myKey")
+    assert sub is not None and sub.size() == W  # same key => same color
+    t = torch.full((10,), float(r + 1))
+    tm.allreduce(t)  # acts on top of stack (the pushed sub-communicator)
+    assert torch.equal(t, torch.full((10,), float(W * (W + 1) / 2)))
+    tm.pop_communicator()
+    t2 = torch.full((4,), 1.0)
+    tm.allreduce(t2)  # back on world
+    assert torch.equal(t2, torch.full((4,), float(W)))
+
+
+def test_push_communicator_string_key():
+    run_distributed(_push_string_key_body, world=3)
