@@ -19,7 +19,8 @@
 // large shards cannot mutually stall (each loop keeps reading while its
 // replies drain). Client enqueues are bounded (backpressure) and fail
 // loudly if the loop has died, so Work.wait() raises instead of hanging.
-// GPU tensors are staged through host memory by the Python wrapper (v1).
+// (This TCP PS serves CPU tensors; GPU tensors use the one-sided xGMI
+// path in ipc_ps.hip and never touch this event loop.)
 #pragma once
 
 #include <atomic>
